@@ -1,0 +1,140 @@
+"""Benchmark harness — the driver contract.
+
+``python bench.py --gpus N --steps K --warmup W`` runs the flagship
+training step (openwebtext_xl 1.5B GPT, bf16, synthetic data, random
+init — BASELINE.json headline) on N GPUs of one node, one rank per GPU
+over RCCL, and prints ONE JSON line from rank 0.
+
+Weak scaling: per-GPU work is fixed at 128 seq x 1024 tok = 131072
+tokens/GPU/step, so N=8 reproduces the reference's global batch 1024
+(reference src/configs/openwebtext_xl.py; BASELINE.md 444K tok/s).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from midgpt_amd.config import load_config
+from midgpt_amd.data import synthetic_batch
+from midgpt_amd.parallel import dist as pdist
+from midgpt_amd.train import build_engine
+from midgpt_amd.utils.lr import warmup_cosine_lr
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--config", default="openwebtext_xl")
+    p.add_argument("--local-batch", type=int, default=128,
+                   help="sequences per GPU per optimizer step")
+    p.add_argument("--micro-batch", type=int, default=None,
+                   help="microbatch size (default: local-batch, i.e. G=1)")
+    args = p.parse_args()
+
+    rank, world, device = pdist.init_distributed()
+    assert world == args.gpus or world == 1, \
+        f"WORLD_SIZE={world} != --gpus {args.gpus}"
+    n = max(world, 1)
+
+    config = load_config(args.config)
+    config.synthetic_data = True
+    config.rundir = ""
+    mc = config.model_config
+    micro = args.micro_batch or args.local_batch
+    assert args.local_batch % micro == 0
+    g_accum = args.local_batch // micro
+    config.batch_size = micro * n
+    config.g_accum_iters = g_accum
+    # ZeRO sharding active whenever world > 1 (headline config shard_model=True)
+
+    torch.manual_seed(1234 + rank)
+    model, engine = build_engine(config, device)
+    tokens_per_step_per_gpu = args.local_batch * mc.block_size
+
+    # pre-generate a couple of synthetic batches on device (data pipeline is
+    # not the measured quantity; the reference benches with real data cached
+    # in RAM — synthetic per the driver contract, no network)
+    batches = []
+    for i in range(2):
+        x, y = synthetic_batch(mc.vocab_size, mc.block_size, micro, g_accum,
+                               device=device)
+        batches.append((x, y))
+
+    def one_step(it):
+        lr = warmup_cosine_lr(it, peak_lr=config.learning_rate,
+                              warmup_steps=config.warmup_steps,
+                              decay_steps=config.lr_decay_steps,
+                              min_lr=config.min_lr)
+        x, y = batches[it % len(batches)]
+        for g in range(g_accum):
+            loss = model.loss(x[g], y[g])
+            loss.backward()
+            engine.microstep_end()
+        engine.step(lr, g_accum)
+        return loss
+
+    for w in range(args.warmup):
+        one_step(w)
+
+    pdist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for it in range(args.steps):
+        loss = one_step(args.warmup + it)
+    pdist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        import torch.distributed as dist
+        td = t.to(device) if device.type == "cuda" else t
+        dist.all_reduce(td, op=dist.ReduceOp.MAX)
+        elapsed = float(td[0])
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    tokens_per_sec = n * tokens_per_step_per_gpu * args.steps / elapsed
+    # model FLOPs/token = 6*N_params + attention term (12*L*D*T per token
+    # ~= 6*L*D*T*2? use 6N + 12*L*T*D); MFU vs 2.5 PF dense bf16 per GPU
+    from midgpt_amd.models.gpt import count_params
+    n_params = count_params(model)
+    flops_per_token = 6 * n_params + 12 * mc.n_layer * mc.n_embd * mc.block_size
+    mfu = tokens_per_sec * flops_per_token / (n * 2.5e15)
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "tokens_per_sec",
+            "value": round(tokens_per_sec, 1),
+            "unit": "tokens/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(tokens_per_sec / 444000.0, 4),
+            "dtype": "bf16" if device.type == "cuda" else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"{args.config}-gpt-{round(n_params/1e6)}M",
+                "global_batch": args.local_batch * n,
+                "seq_len": mc.block_size,
+                "parallelism": (f"zero-dp{n}" if engine.zero else f"dp{n}"),
+                "g_accum_iters": g_accum,
+                "mfu_vs_2.5pf_dense": round(mfu, 4),
+                "last_loss": round(float(loss), 4),
+            },
+        }), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,445 @@
+// Flash-style causal attention, forward + backward (plans K1/K2).
+// MI355X-native: MFMA v_mfma_f32_32x32x16_bf16 tiles, LDS-staged K/V with
+// XOR swizzles (guide T2/G4), online fp32 softmax held IN REGISTERS via the
+// swapped-QK^T layout + permlane32_swap half exchanges (guide T12), fp32
+// LSE saved for the recompute-based backward.
+//
+// Reference numerics: softmax(mask(QK^T)/sqrt(C)) in fp32, bf16 elsewhere
+// (reference src/model.py:71-79). The T x T score matrix is never
+// materialized.
+//
+// Geometry (fwd): one workgroup = 4 waves = 128 q rows (32/wave);
+// KV tiles of 32 rows double-buffered in LDS; grid = B*H*(T/128).
+// Geometry (bwd): one workgroup = 4 waves = 128 k rows; iterates q tiles;
+// dK/dV accumulate in registers, dQ via fp32 atomics (v1; split-q kernel
+// is the planned v2).
+#include "common.h"
+#include "mfma.h"
+
+// ---------------------------------------------------------------------------
+// LDS swizzles
+// ---------------------------------------------------------------------------
+// Row-major [R][C] bf16 tile, rows of C*2 bytes, read by ds_read_b128 at
+// per-lane rows: XOR byte bits 4..7 with row (C=128: row&15 -> conflict-free;
+// C=64 rows are 128 B: row&7).
+template <int C> DEVINL int swz_rm(int row, int byte_in_row) {
+  constexpr int M = (C == 128) ? 15 : 7;
+  return byte_in_row ^ ((row & M) << 4);
+}
+// Transposed [C][32] bf16 tile (rows of 64 B): XOR byte bits 4..5 with
+// (row>>2)&3 (see analysis: removes the 4-way conflict of the 64-B stride).
+DEVINL int swz_tr(int row, int byte_in_row) {
+  return byte_in_row ^ (((row >> 2) & 3) << 4);
+}
+
+// ---------------------------------------------------------------------------
+// Cooperative staging helpers (256 threads)
+// ---------------------------------------------------------------------------
+// Row-major 32 x C tile from global (row stride C) into swizzled LDS.
+template <int C>
+DEVINL void stage_rm(const u16* __restrict__ g, u16* lds) {
+#pragma unroll
+  for (int idx = threadIdx.x * 8; idx < 32 * C; idx += 256 * 8) {
+    const int row = idx / C, col = idx % C;
+    u16x8 val = *(const u16x8*)(g + row * C + col);
+    *(u16x8*)((char*)lds + row * C * 2 + swz_rm<C>(row, col * 2)) = val;
+  }
+}
+// Transposed: global 32 x C (row stride C) -> LDS [C][32] swizzled.
+template <int C>
+DEVINL void stage_tr(const u16* __restrict__ g, u16* lds) {
+#pragma unroll
+  for (int idx = threadIdx.x * 8; idx < 32 * C; idx += 256 * 8) {
+    const int row = idx / C, col = idx % C;  // row = kv/q index, col = c
+    u16x8 val = *(const u16x8*)(g + row * C + col);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int r = col + j;  // LDS row = c
+      *(u16*)((char*)lds + r * 64 + swz_tr(r, row * 2)) = val[j];
+    }
+  }
+}
+
+// Read one A/B fragment (bf16x8) from a swizzled row-major [R][C] tile:
+// lane l -> row (l&31)+row0, bytes 16*chunk32? caller passes byte base.
+template <int C>
+DEVINL bf16x8_t read_rm_frag(const u16* lds, int row, int cbyte) {
+  return *(const bf16x8_t*)((const char*)lds + row * C * 2 + swz_rm<C>(row, cbyte));
+}
+DEVINL bf16x8_t read_tr_frag(const u16* lds, int row, int kbyte) {
+  return *(const bf16x8_t*)((const char*)lds + row * 64 + swz_tr(row, kbyte));
+}
+
+DEVINL float shfl32(float v, int src) { return __shfl(v, src, 32); }
+
+// ===========================================================================
+// Forward
+// ===========================================================================
+template <int C>
+__global__ void attn_fwd_kernel(const u16* __restrict__ q,
+                                const u16* __restrict__ k,
+                                const u16* __restrict__ v,
+                                u16* __restrict__ o, float* __restrict__ lse,
+                                int B, int H, int T) {
+  constexpr int NCB = C / 32;   // 32-col c-blocks
+  constexpr int NCH = C / 16;   // 16-deep mfma chunks
+  const float scale = rsqrtf((float)C);
+  const int nQB = T / 128;
+  const int qb = blockIdx.x % nQB;
+  const long bh = blockIdx.x / nQB;  // b*H + h
+  const int q0 = qb * 128;
+  const int lane = lane_id();
+  const int w = wave_id();
+  const int qw0 = q0 + 32 * w;          // this wave's first q row
+  const int myq = qw0 + (lane & 31);    // this lane's q row
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* ldsK = (u16*)smem;                       // [2][32*C]
+  u16* ldsVt = (u16*)(smem + 2 * 32 * C * 2);   // [2][C*32]
+  float* obuf = (float*)smem;                   // epilogue reuse: [4][32*32]
+
+  const u16* qg = q + (bh * T) * C;
+  const u16* kg = k + (bh * T) * C;
+  const u16* vg = v + (bh * T) * C;
+
+  // Q B-fragments straight from global to registers.
+  bf16x8_t qf[NCH];
+  {
+    const u16* qrow = qg + (long)myq * C;
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      qf[ch] = *(const bf16x8_t*)(qrow + 16 * ch + 8 * (lane >> 5));
+  }
+
+  f32x16 oacc[NCB];
+#pragma unroll
+  for (int cb = 0; cb < NCB; ++cb) oacc[cb] = (f32x16)(0.f);
+  float m = -1e30f, lsum = 0.f;
+
+  const int nkt = (q0 + 128) / 32;
+  stage_rm<C>(kg, ldsK);
+  stage_tr<C>(vg, ldsVt);
+  __syncthreads();
+
+  for (int kt = 0; kt < nkt; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < nkt) {
+      stage_rm<C>(kg + (long)(kt + 1) * 32 * C, ldsK + (1 - buf) * 32 * C);
+      stage_tr<C>(vg + (long)(kt + 1) * 32 * C, ldsVt + (1 - buf) * C * 32);
+    }
+    const int k0 = kt * 32;
+    if (k0 <= qw0 + 31) {  // wave-uniform: tile not fully masked for this wave
+      // S = K x Q^T  (swapped: D rows = k, cols = q)
+      f32x16 s = (f32x16)(0.f);
+      const u16* kb = ldsK + buf * 32 * C;
+#pragma unroll
+      for (int ch = 0; ch < NCH; ++ch) {
+        bf16x8_t a = read_rm_frag<C>(kb, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+        s = mfma_32x32x16_bf16(a, qf[ch], s);
+      }
+      // mask + tile row-max (per q = lane&31; halves merged via xor 32)
+      float sv[16];
+      float mt = -1e30f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        sv[r] = (k0 + mfma_d_row(lane, r) > myq) ? -1e30f : s[r];
+        mt = fmaxf(mt, sv[r]);
+      }
+      mt = fmaxf(mt, __shfl_xor(mt, 32));
+      const float mn = fmaxf(m, mt);
+      const float alpha = __expf((m - mn) * scale);
+      m = mn;
+      float p[16], psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        p[r] = __expf((sv[r] - mn) * scale);
+        psum += p[r];
+      }
+      psum += __shfl_xor(psum, 32);
+      lsum = lsum * alpha + psum;
+      // rescale O by alpha[q_of_reg]
+      float arow[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) arow[r] = shfl32(alpha, mfma_d_row(lane, r));
+#pragma unroll
+      for (int cb = 0; cb < NCB; ++cb)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[cb][r] *= arow[r];
+      // P -> bf16 A-fragments; PV
+      bf16x8_t pf0 = dlayout_to_afrag(p);
+      bf16x8_t pf1 = dlayout_to_afrag(p + 8);
+      const u16* vb = ldsVt + buf * C * 32;
+#pragma unroll
+      for (int cb = 0; cb < NCB; ++cb) {
+        bf16x8_t b0 = read_tr_frag(vb, 32 * cb + (lane & 31), 16 * (lane >> 5));
+        bf16x8_t b1 = read_tr_frag(vb, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+        oacc[cb] = mfma_32x32x16_bf16(pf0, b0, oacc[cb]);
+        oacc[cb] = mfma_32x32x16_bf16(pf1, b1, oacc[cb]);
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: normalize, bounce through LDS, wide stores
+  const float rec = 1.f / lsum;
+  float rrow[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) rrow[r] = shfl32(rec, mfma_d_row(lane, r));
+  if (lane < 32) lse[bh * T + myq] = m * scale + __logf(lsum);
+  float* ob = obuf + w * 32 * 32;
+  u16* og = o + (bh * T + qw0) * C;
+#pragma unroll
+  for (int cb = 0; cb < NCB; ++cb) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = mfma_d_row(lane, r);
+      *(float*)((char*)ob + row * 128 + (((lane & 31) * 4) ^ ((row & 7) << 4))) =
+          oacc[cb][r] * rrow[r];
+    }
+    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt: LDS writes land (same wave)
+    const int row = lane & 31;
+    const int c16 = 16 * (lane >> 5);
+    float tmp[16];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      f32x4 t = *(const f32x4*)((char*)ob + row * 128 + (((c16 + 4 * i) * 4) ^ ((row & 7) << 4)));
+      tmp[4 * i] = t[0]; tmp[4 * i + 1] = t[1]; tmp[4 * i + 2] = t[2]; tmp[4 * i + 3] = t[3];
+    }
+    u16x8 out0, out1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { out0[j] = f2b(tmp[j]); out1[j] = f2b(tmp[8 + j]); }
+    *(u16x8*)(og + (long)row * C + 32 * cb + c16) = out0;
+    *(u16x8*)(og + (long)row * C + 32 * cb + c16 + 8) = out1;
+    __builtin_amdgcn_s_waitcnt(0);
+  }
+}
+
+// ===========================================================================
+// delta = rowsum(dO * O) — one wave per row (prologue of backward)
+// ===========================================================================
+__global__ void attn_delta_kernel(const u16* __restrict__ dO,
+                                  const u16* __restrict__ O,
+                                  float* __restrict__ delta, long N, int C) {
+  const int lane = lane_id();
+  const long row0 = (long)blockIdx.x * (blockDim.x / WAVE) + wave_id();
+  const long rstep = (long)gridDim.x * (blockDim.x / WAVE);
+  for (long row = row0; row < N; row += rstep) {
+    float acc = 0.f;
+    for (int i = lane * 8; i < C; i += WAVE * 8) {
+      u16x8 a = *(const u16x8*)(dO + row * C + i);
+      u16x8 b = *(const u16x8*)(O + row * C + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc += b2f(a[j]) * b2f(b[j]);
+    }
+    acc = group_sum<WAVE>(acc);
+    if (lane == 0) delta[row] = acc;
+  }
+}
+
+// ===========================================================================
+// Backward: one WG = 128 k rows (32/wave); iterates q tiles >= diagonal.
+// dK/dV in registers; dQ via fp32 atomics. S recomputed from Q,K + LSE.
+// ===========================================================================
+template <int C>
+__global__ void attn_bwd_kernel(const u16* __restrict__ dO,
+                                const u16* __restrict__ q,
+                                const u16* __restrict__ k,
+                                const u16* __restrict__ v,
+                                const float* __restrict__ lse,
+                                const float* __restrict__ delta,
+                                float* __restrict__ dq32,
+                                u16* __restrict__ dk,
+                                u16* __restrict__ dv,
+                                int B, int H, int T) {
+  constexpr int NCB = C / 32;
+  constexpr int NCH = C / 16;
+  const float scale = rsqrtf((float)C);
+  const int nKB = T / 128;
+  const int kb = blockIdx.x % nKB;
+  const long bh = blockIdx.x / nKB;
+  const int lane = lane_id();
+  const int w = wave_id();
+  const int kw0 = kb * 128 + 32 * w;    // wave's first k row
+  const int myk = kw0 + (lane & 31);    // lane's k row
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // [Qrm 32*C | Qt C*32 | dOrm 32*C | dOt C*32 | Kt 4*(C*32) | dS 4*(32*32)
+  //  | lse 32 | delta 32 ]  (bf16 tiles; dS bf16)
+  u16* ldsQ = (u16*)smem;
+  u16* ldsQt = ldsQ + 32 * C;
+  u16* ldsDO = ldsQt + C * 32;
+  u16* ldsDOt = ldsDO + 32 * C;
+  u16* ldsKt = ldsDOt + C * 32;          // per wave: + w*C*32
+  u16* ldsDS = ldsKt + 4 * C * 32;       // per wave: + w*32*32
+  float* ldsLse = (float*)(ldsDS + 4 * 32 * 32);
+  float* ldsDelta = ldsLse + 32;
+
+  const u16* qg = q + (bh * T) * C;
+  const u16* kg = k + (bh * T) * C;
+  const u16* vg = v + (bh * T) * C;
+  const u16* dog = dO + (bh * T) * C;
+
+  // wave-owned K and V row fragments (B-fragment shape: contiguous 8 c)
+  bf16x8_t kf[NCH], vf[NCH];
+  {
+    const u16* krow = kg + (long)myk * C;
+    const u16* vrow = vg + (long)myk * C;
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch) {
+      kf[ch] = *(const bf16x8_t*)(krow + 16 * ch + 8 * (lane >> 5));
+      vf[ch] = *(const bf16x8_t*)(vrow + 16 * ch + 8 * (lane >> 5));
+    }
+  }
+  // wave-private transposed K image for the dQ mfma (B operand needs
+  // column access): Kt[c][k_local 0..31]
+  {
+    u16* kt = ldsKt + w * C * 32;
+    for (int idx = lane * 8; idx < 32 * C; idx += WAVE * 8) {
+      const int row = idx / C, col = idx % C;
+      u16x8 val = *(const u16x8*)(kg + (long)(kw0 + row) * C + col);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *(u16*)((char*)(kt + (col + j) * 32) + swz_tr(col + j, row * 2)) = val[j];
+    }
+  }
+
+  f32x16 dvacc[NCB], dkacc[NCB];
+#pragma unroll
+  for (int cb = 0; cb < NCB; ++cb) { dvacc[cb] = (f32x16)(0.f); dkacc[cb] = (f32x16)(0.f); }
+
+  const int qt0 = kb * 128 / 32;  // first q tile (diagonal)
+  const int nqt = T / 32;
+  for (int qt = qt0; qt < nqt; ++qt) {
+    const int qbase = qt * 32;
+    __syncthreads();
+    stage_rm<C>(qg + (long)qbase * C, ldsQ);
+    stage_tr<C>(qg + (long)qbase * C, ldsQt);
+    stage_rm<C>(dog + (long)qbase * C, ldsDO);
+    stage_tr<C>(dog + (long)qbase * C, ldsDOt);
+    if (threadIdx.x < 32) {
+      ldsLse[threadIdx.x] = lse[bh * T + qbase + threadIdx.x];
+      ldsDelta[threadIdx.x] = delta[bh * T + qbase + threadIdx.x];
+    }
+    __syncthreads();
+    if (qbase + 31 < kw0) continue;  // fully masked for this wave
+
+    // S = Q x K^T  (D rows = q regs, cols = k lanes)
+    f32x16 s = (f32x16)(0.f);
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch) {
+      bf16x8_t a = read_rm_frag<C>(ldsQ, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+      s = mfma_32x32x16_bf16(a, kf[ch], s);
+    }
+    // P = exp(s*scale - lse[q]), masked
+    float p[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = qbase + mfma_d_row(lane, r);
+      p[r] = (myk > qrow) ? 0.f
+           : __expf(s[r] * scale - ldsLse[mfma_d_row(lane, r)]);
+    }
+    // dV += P^T x dO   (A = packed P: rows k, kdim q)
+    bf16x8_t pf0 = dlayout_to_afrag(p);
+    bf16x8_t pf1 = dlayout_to_afrag(p + 8);
+#pragma unroll
+    for (int cb = 0; cb < NCB; ++cb) {
+      bf16x8_t b0 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 16 * (lane >> 5));
+      bf16x8_t b1 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+      dvacc[cb] = mfma_32x32x16_bf16(pf0, b0, dvacc[cb]);
+      dvacc[cb] = mfma_32x32x16_bf16(pf1, b1, dvacc[cb]);
+    }
+    // dP = dO x V^T  (D rows = q regs, cols = k lanes)
+    f32x16 dp = (f32x16)(0.f);
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch) {
+      bf16x8_t a = read_rm_frag<C>(ldsDO, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+      dp = mfma_32x32x16_bf16(a, vf[ch], dp);
+    }
+    // dS = P * (dP - delta[q]) * scale  (keep layout)
+    float ds[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      ds[r] = p[r] * (dp[r] - ldsDelta[mfma_d_row(lane, r)]) * scale;
+    // dK += dS^T x Q  (A = packed dS: rows k, kdim q; B = Qt)
+    bf16x8_t df0 = dlayout_to_afrag(ds);
+    bf16x8_t df1 = dlayout_to_afrag(ds + 8);
+#pragma unroll
+    for (int cb = 0; cb < NCB; ++cb) {
+      bf16x8_t b0 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 16 * (lane >> 5));
+      bf16x8_t b1 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+      dkacc[cb] = mfma_32x32x16_bf16(df0, b0, dkacc[cb]);
+      dkacc[cb] = mfma_32x32x16_bf16(df1, b1, dkacc[cb]);
+    }
+    // dQ: transpose dS through per-wave LDS, then dQ += dS x K
+    {
+      u16* dsl = ldsDS + w * 32 * 32;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = mfma_d_row(lane, r);  // q-local
+        *(u16*)((char*)(dsl + row * 32) + swz_tr(row, (lane & 31) * 2)) = f2b(ds[r]);
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+      f32x16 dqa[NCB];
+#pragma unroll
+      for (int cb = 0; cb < NCB; ++cb) dqa[cb] = (f32x16)(0.f);
+      const u16* kt = ldsKt + w * C * 32;
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        // A[q = lane&31][k = 16*kc + 8*(lane>>5) + j]
+        bf16x8_t a = *(const bf16x8_t*)((const char*)(dsl + (lane & 31) * 32) +
+                               swz_tr(lane & 31, (16 * kc + 8 * (lane >> 5)) * 2));
+#pragma unroll
+        for (int cb = 0; cb < NCB; ++cb) {
+          bf16x8_t b = read_tr_frag(kt, 32 * cb + (lane & 31),
+                                    (16 * kc + 8 * (lane >> 5)) * 2);
+          dqa[cb] = mfma_32x32x16_bf16(a, b, dqa[cb]);
+        }
+      }
+#pragma unroll
+      for (int cb = 0; cb < NCB; ++cb)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qrow = qbase + mfma_d_row(lane, r);
+          atomicAdd(dq32 + ((bh * T + qrow) * C + 32 * cb + (lane & 31)),
+                    dqa[cb][r]);
+        }
+    }
+  }
+
+  // epilogue: dK, dV regs -> LDS bounce -> wide bf16 stores
+  __syncthreads();
+  float* ob = (float*)smem + w * 32 * 32;
+  u16* dkg = dk + (bh * T + kw0) * C;
+  u16* dvg = dv + (bh * T + kw0) * C;
+#pragma unroll
+  for (int which = 0; which < 2; ++which) {
+    f32x16* acc = which == 0 ? dkacc : dvacc;
+    u16* out = which == 0 ? dkg : dvg;
+#pragma unroll
+    for (int cb = 0; cb < NCB; ++cb) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = mfma_d_row(lane, r);
+        *(float*)((char*)ob + row * 128 + (((lane & 31) * 4) ^ ((row & 7) << 4))) =
+            acc[cb][r];
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+      const int row = lane & 31;
+      const int c16 = 16 * (lane >> 5);
+      float tmp[16];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        f32x4 t = *(const f32x4*)((char*)ob + row * 128 +
+                                  (((c16 + 4 * i) * 4) ^ ((row & 7) << 4)));
+        tmp[4 * i] = t[0]; tmp[4 * i + 1] = t[1];
+        tmp[4 * i + 2] = t[2]; tmp[4 * i + 3] = t[3];
+      }
+      u16x8 o0, o1;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) { o0[j] = f2b(tmp[j]); o1[j] = f2b(tmp[8 + j]); }
+      *(u16x8*)(out + (long)row * C + 32 * cb + c16) = o0;
+      *(u16x8*)(out + (long)row * C + 32 * cb + c16 + 8) = o1;
+      __builtin_amdgcn_s_waitcnt(0);
+    }
+  }
+}

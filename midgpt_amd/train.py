@@ -1,0 +1,161 @@
+"""Training runtime: train(config) — the MI355X-native equivalent of the
+reference's src/train.py:127-225.
+
+Flow parity (reference call stack):
+  init distributed -> data load / per-process split -> engine (flat bf16
+  weights + ZeRO fp32 master) -> restore latest checkpoint -> hot loop
+  { eval every eval_interval | get_batch -> H2D -> G microsteps
+    (fwd+bwd with per-block remat, bf16 compute, reduce-scatter per
+    microstep) -> fused AdamW step -> log } -> final checkpoint wait.
+"""
+from __future__ import annotations
+
+import json
+import os
+import time
+
+import torch
+
+from midgpt_amd import ops
+from midgpt_amd.config import ExperimentConfig
+from midgpt_amd.data import BatchLoader
+from midgpt_amd.models.gpt import GPT, count_params
+from midgpt_amd.parallel import dist as pdist
+from midgpt_amd.parallel.engine import ShardedAdamW
+from midgpt_amd.utils import checkpoint as ckpt
+from midgpt_amd.utils.lr import warmup_cosine_lr
+
+DTYPES = {"float32": torch.float32, "bfloat16": torch.bfloat16}
+
+
+def build_engine(config: ExperimentConfig, device: torch.device,
+                 generator: torch.Generator | None = None
+                 ) -> tuple[GPT, ShardedAdamW]:
+    model = GPT(config.model_config, generator)
+    model = model.to(device)
+    model.remat = config.remat
+    compute_dtype = DTYPES[config.compute_dtype]
+    if device.type == "cpu" and compute_dtype == torch.bfloat16:
+        # CPU path runs fp32 (bf16 CPU matmuls are slow and the CPU tier is
+        # a correctness tier); GPU honors the config.
+        compute_dtype = torch.float32
+    engine = ShardedAdamW(
+        model, compute_dtype=compute_dtype, zero=config.shard_model,
+        beta1=config.beta1, beta2=config.beta2, eps=config.adam_eps,
+        weight_decay=config.weight_decay, peak_lr=config.learning_rate,
+        grad_clip=config.grad_clip, device=device)
+    return model, engine
+
+
+@torch.no_grad()
+def evaluate(model: GPT, loader: BatchLoader, batch_size: int,
+             device: torch.device, n_batches: int = 200) -> float:
+    """Mean loss over fixed-size batches (reference src/train.py:107-117)."""
+    model.eval()
+    total = 0.0
+    for _ in range(n_batches):
+        x, y = loader.batch("val", batch_size, 1)
+        x, y = x[0].to(device), y[0].to(device)
+        total += float(model.loss(x, y))
+    model.train()
+    return total / n_batches
+
+
+def train(config: ExperimentConfig):
+    rank, world, device = pdist.init_distributed()
+    torch.manual_seed(1234 + rank if config.seed is None else config.seed + rank)
+    gen = torch.Generator().manual_seed(config.seed) if config.seed is not None else None
+
+    model, engine = build_engine(config, device, gen)
+    if pdist.is_main():
+        print(f"params: {count_params(model)/1e6:.1f}M  world={world} "
+              f"device={device} zero={engine.zero}")
+
+    assert config.batch_size % world == 0, "global batch must divide world size"
+    local_bs = config.batch_size // world
+
+    loader = BatchLoader(config.data_dir, config.model_config.vocab_size,
+                         config.model_config.block_size, rank, world,
+                         synthetic=config.synthetic_data, seed=config.seed)
+
+    mngr = None
+    first_step = 0
+    if config.rundir:
+        mngr = ckpt.CheckpointManager(
+            config.rundir, 0 if config.debug else config.eval_interval)
+        state = ckpt.load_full_state(config.rundir)
+        if state is not None:
+            engine.load_state_full(state["master"], state["m"], state["v"],
+                                   state["step_count"])
+            first_step = state["step"] + 1
+            if pdist.is_main():
+                print(f"resumed from step {state['step']}")
+
+    eval_batches = 1 if config.debug else 200
+    t0 = time.perf_counter()
+    for it in range(first_step, config.max_steps):
+        if it % config.eval_interval == 0:
+            tl = evaluate(model, loader, local_bs, device, eval_batches)
+            vl = evaluate(model, loader, local_bs, device, eval_batches)
+            if pdist.is_main():
+                print(f"step {it}: loss/train {tl:.4f} loss/val {vl:.4f}")
+            log_metrics(config, it, {"loss/train": tl, "loss/val": vl})
+
+        lr = warmup_cosine_lr(it, peak_lr=config.learning_rate,
+                              warmup_steps=config.warmup_steps,
+                              decay_steps=config.lr_decay_steps,
+                              min_lr=config.min_lr)
+        x, y = loader.batch("train", local_bs, config.g_accum_iters)
+        loss_sum = 0.0
+        for g in range(config.g_accum_iters):
+            xg, yg = x[g].to(device, non_blocking=True), y[g].to(device, non_blocking=True)
+            loss = model.loss(xg, yg)
+            loss.backward()
+            engine.microstep_end()
+            loss_sum += float(loss)
+        engine.step(lr, config.g_accum_iters)
+        loss_step = loss_sum / config.g_accum_iters
+
+        if it % 20 == 0:
+            log_metrics(config, it, {"loss/optimized": loss_step})
+        if mngr is not None and mngr.should_save(it):
+            mngr.save(it, engine)
+        if pdist.is_main() and it % 10 == 0:
+            dt = time.perf_counter() - t0
+            thpt = (it - first_step + 1) * config.batch_size * \
+                config.g_accum_iters / max(dt, 1e-9)
+            print(f"step {it}: loss {loss_step:.4f} lr {lr:.2e} "
+                  f"thpt {thpt:.1f} seq/s", flush=True)
+    if mngr is not None:
+        mngr.save(config.max_steps - 1, engine)
+        mngr.wait()
+
+
+_WANDB = None
+
+
+def log_metrics(config: ExperimentConfig, step: int, metrics: dict):
+    """wandb if available and configured (proc 0 only), else no-op."""
+    global _WANDB
+    if not pdist.is_main():
+        return
+    if _WANDB is None:
+        _WANDB = False
+        if os.environ.get("WANDB_API_KEY"):
+            try:
+                import wandb
+                run_id = None
+                idfile = os.path.join(config.rundir, "wandb_id.txt")
+                if config.rundir and os.path.exists(idfile):
+                    run_id = open(idfile).read().strip()
+                r = wandb.init(project="midgpt", id=run_id, resume="allow",
+                               config=json.loads(config.to_json()))
+                if config.rundir:
+                    with open(idfile, "w") as f:
+                        f.write(r.id)
+                _WANDB = True
+            except Exception:
+                _WANDB = False
+    if _WANDB:
+        import wandb
+        wandb.log(metrics, step=step)

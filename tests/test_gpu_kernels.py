@@ -1,0 +1,169 @@
+"""GPU numerics tests: each HIP kernel vs the plain-PyTorch fp32 reference.
+All marked gpu (run on MI355X via gpurun / the driver)."""
+import math
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from midgpt_amd import ops
+    from midgpt_amd.ops import reference as ref
+    assert ops.have_ext(), f"HIP extension must be built: {ops._C_ERR}"
+
+DEV = "cuda:0"
+
+
+def relerr(a, b):
+    a, b = a.float(), b.float()
+    return float((a - b).norm() / (b.norm() + 1e-12))
+
+
+# ---------------------------------------------------------------------------
+def test_mfma_layout_probe():
+    """Verifies the A/B/D fragment-layout maps in mfma.h on hardware."""
+    torch.manual_seed(0)
+    A = torch.randn(32, 16, device=DEV)
+    B = torch.randn(16, 32, device=DEV)  # asymmetric: catches transposes
+    D = ops._C.probe_mfma(A.contiguous(), B.contiguous())
+    expect = (A.to(torch.bfloat16).float() @ B.to(torch.bfloat16).float())
+    assert relerr(D, expect) < 2e-2, (D - expect).abs().max()
+
+
+def test_mfma_pack_probe():
+    """Verifies dlayout_to_afrag (cvt_pk + permlane32_swap path): M enters in
+    D-layout, result must be M^T @ B."""
+    torch.manual_seed(1)
+    M = torch.randn(32, 32, device=DEV)
+    B = torch.randn(32, 32, device=DEV)
+    D = ops._C.probe_pack(M.contiguous(), B.contiguous())
+    expect = M.t().to(torch.bfloat16).float() @ B.to(torch.bfloat16).float()
+    assert relerr(D, expect) < 2e-2, (D - expect).abs().max()
+
+
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("N,D", [(128, 768), (64, 2048), (32, 4096)])
+def test_rmsnorm_fwd_bwd(N, D):
+    torch.manual_seed(2)
+    x = torch.randn(N, D, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    y = ops.rmsnorm(x, None, 1e-6)
+    y_ref = ref.rmsnorm(x.detach().float(), None, 1e-6)
+    assert relerr(y, y_ref) < 1e-2
+    g = torch.randn_like(y)
+    (y.float() * g.float()).sum().backward()
+    x2 = x.detach().float().requires_grad_(True)
+    y2 = x2 * torch.rsqrt(x2.pow(2).mean(-1, keepdim=True) + 1e-6)
+    (y2 * g.float()).sum().backward()
+    assert relerr(x.grad, x2.grad) < 2e-2
+
+
+@pytest.mark.parametrize("B,T,H,C", [(2, 128, 4, 64), (1, 256, 2, 128)])
+def test_qkv_prep_fwd_bwd(B, T, H, C):
+    torch.manual_seed(3)
+    qkv = torch.randn(B, T, 3, H, C, device=DEV, dtype=torch.bfloat16,
+                      requires_grad=True)
+    qw = torch.randn(C, device=DEV, requires_grad=True)
+    kw = torch.randn(C, device=DEV, requires_grad=True)
+    sin, cos = ref.rope_tables(C, T, device=DEV)
+    q, k, v = ops.qkv_prep(qkv, qw, kw, sin, cos)
+    # reference on fp32 copies
+    qr = ref.qk_layernorm(qkv.detach().float()[:, :, 0].permute(0, 2, 1, 3), qw.detach())
+    kr = ref.qk_layernorm(qkv.detach().float()[:, :, 1].permute(0, 2, 1, 3), kw.detach())
+    qe = ref.apply_rope(qr, sin, cos)
+    ke = ref.apply_rope(kr, sin, cos)
+    ve = qkv.detach().float()[:, :, 2].permute(0, 2, 1, 3)
+    assert relerr(q, qe) < 1e-2, relerr(q, qe)
+    assert relerr(k, ke) < 1e-2
+    assert relerr(v, ve) < 1e-3
+    # backward
+    gq, gk, gv = torch.randn_like(q), torch.randn_like(k), torch.randn_like(v)
+    (q.float() * gq.float() + k.float() * gk.float() + v.float() * gv.float()).sum().backward()
+    qkv2 = qkv.detach().float().requires_grad_(True)
+    qw2 = qw.detach().clone().requires_grad_(True)
+    kw2 = kw.detach().clone().requires_grad_(True)
+    qr2 = ref.qk_layernorm(qkv2[:, :, 0].permute(0, 2, 1, 3), qw2)
+    kr2 = ref.qk_layernorm(qkv2[:, :, 1].permute(0, 2, 1, 3), kw2)
+    qe2 = ref.apply_rope(qr2, sin, cos)
+    ke2 = ref.apply_rope(kr2, sin, cos)
+    ve2 = qkv2[:, :, 2].permute(0, 2, 1, 3)
+    (qe2 * gq.float() + ke2 * gk.float() + ve2 * gv.float()).sum().backward()
+    assert relerr(qkv.grad, qkv2.grad) < 2e-2
+    assert relerr(qw.grad, qw2.grad) < 2e-2
+    assert relerr(kw.grad, kw2.grad) < 2e-2
+
+
+@pytest.mark.parametrize("B,H,T,C", [(2, 2, 128, 64), (1, 2, 256, 128),
+                                     (2, 1, 1024, 128)])
+def test_attention_fwd(B, H, T, C):
+    torch.manual_seed(4)
+    q = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16)
+    o, lse = ops._C.attn_fwd(q, k, v)
+    # fp32 reference on the same (bf16-rounded) inputs
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) / math.sqrt(C)
+    mask = torch.ones(T, T, dtype=torch.bool, device=DEV).tril()
+    s = s.masked_fill(~mask, float("-inf"))
+    lse_ref = torch.logsumexp(s, dim=-1)
+    o_ref = torch.matmul(torch.softmax(s, -1), v.float())
+    assert relerr(lse, lse_ref) < 1e-3, relerr(lse, lse_ref)
+    assert relerr(o, o_ref) < 2e-2, relerr(o, o_ref)
+
+
+@pytest.mark.parametrize("B,H,T,C", [(2, 2, 128, 64), (1, 2, 256, 128)])
+def test_attention_bwd(B, H, T, C):
+    torch.manual_seed(5)
+    q = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    o = ops.flash_attention(q, k, v)
+    do = torch.randn_like(o)
+    (o.float() * do.float()).sum().backward()
+    # fp32 reference
+    q2, k2, v2 = (t.detach().float().requires_grad_(True) for t in (q, k, v))
+    s = torch.matmul(q2, k2.transpose(-1, -2)) / math.sqrt(C)
+    mask = torch.ones(T, T, dtype=torch.bool, device=DEV).tril()
+    s = s.masked_fill(~mask, float("-inf"))
+    o2 = torch.matmul(torch.softmax(s, -1), v2)
+    (o2 * do.float()).sum().backward()
+    assert relerr(q.grad, q2.grad) < 4e-2, ("dq", relerr(q.grad, q2.grad))
+    assert relerr(k.grad, k2.grad) < 4e-2, ("dk", relerr(k.grad, k2.grad))
+    assert relerr(v.grad, v2.grad) < 4e-2, ("dv", relerr(v.grad, v2.grad))
+
+
+def test_cross_entropy_fwd_bwd():
+    torch.manual_seed(6)
+    N, V = 512, 50304
+    logits = (torch.randn(N, V, device=DEV) * 2).to(torch.bfloat16).requires_grad_(True)
+    targets = torch.randint(0, V, (N,), device=DEV)
+    loss = ops.cross_entropy(logits, targets)
+    l2 = logits.detach().float().requires_grad_(True)
+    expect = torch.nn.functional.cross_entropy(l2, targets)
+    assert abs(float(loss) - float(expect)) < 1e-3 * float(expect)
+    loss.backward()
+    expect.backward()
+    assert relerr(logits.grad, l2.grad) < 2e-2
+
+
+def test_adamw_step_gpu_matches_cpu():
+    torch.manual_seed(7)
+    n = 100003
+    master = torch.randn(n, device=DEV)
+    grad = torch.randn(n, device=DEV)
+    m = torch.randn(n, device=DEV).abs() * 0.1
+    v = torch.randn(n, device=DEV).abs() * 0.01
+    out16 = torch.empty(n, device=DEV, dtype=torch.bfloat16)
+    cm, cg = master.cpu().clone(), grad.cpu().clone()
+    cmm, cv = m.cpu().clone(), v.cpu().clone()
+    sq = (grad * grad).sum()
+    kw = dict(lr=1e-3, beta1=0.9, beta2=0.95, eps=1e-8, wd_over_peak_lr=0.1,
+              grad_scale=0.125, clip_norm=1.0, step=3)
+    ops.adamw_step(master, grad, m, v, out16, sq_sum=sq, **kw)
+    co = torch.empty(n, dtype=torch.bfloat16)
+    ops.adamw_step(cm, cg, cmm, cv, co, sq_sum=sq.cpu(), **kw)
+    assert relerr(master.cpu(), cm) < 1e-6
+    assert relerr(m.cpu(), cmm) < 1e-6
+    assert relerr(v.cpu(), cv) < 1e-6
+    assert torch.equal(out16.cpu(), co)
