@@ -76,7 +76,7 @@ DEVINL float shfl32(float v, int src) { return __shfl(v, src, 32); }
 // Forward
 // ===========================================================================
 template <int C>
-__global__ void attn_fwd_kernel(const u16* __restrict__ q,
+__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(const u16* __restrict__ q,
                                 const u16* __restrict__ k,
                                 const u16* __restrict__ v,
                                 u16* __restrict__ o, float* __restrict__ lse,
@@ -241,7 +241,7 @@ __global__ void attn_delta_kernel(const u16* __restrict__ dO,
 // dK/dV in registers; dQ via fp32 atomics. S recomputed from Q,K + LSE.
 // ===========================================================================
 template <int C>
-__global__ void attn_bwd_kernel(const u16* __restrict__ dO,
+__global__ __launch_bounds__(256, 1) void attn_bwd_kernel(const u16* __restrict__ dO,
                                 const u16* __restrict__ q,
                                 const u16* __restrict__ k,
                                 const u16* __restrict__ v,
@@ -279,16 +279,16 @@ __global__ void attn_bwd_kernel(const u16* __restrict__ dO,
   const u16* vg = v + (bh * T) * C;
   const u16* dog = dO + (bh * T) * C;
 
-  // wave-owned K and V row fragments (B-fragment shape: contiguous 8 c)
-  bf16x8_t kf[NCH], vf[NCH];
+  // wave-owned K row fragments (B-fragment shape: contiguous 8 c).
+  // V fragments are re-read from global per q-tile (L2-resident) to keep
+  // register pressure under the 256-VGPR spill cliff.
+  bf16x8_t kf[NCH];
+  const u16* vrow = vg + (long)myk * C;
   {
     const u16* krow = kg + (long)myk * C;
-    const u16* vrow = vg + (long)myk * C;
 #pragma unroll
-    for (int ch = 0; ch < NCH; ++ch) {
+    for (int ch = 0; ch < NCH; ++ch)
       kf[ch] = *(const bf16x8_t*)(krow + 16 * ch + 8 * (lane >> 5));
-      vf[ch] = *(const bf16x8_t*)(vrow + 16 * ch + 8 * (lane >> 5));
-    }
   }
   // wave-private transposed K image for the dQ mfma (B operand needs
   // column access): Kt[c][k_local 0..31]
@@ -353,7 +353,8 @@ __global__ void attn_bwd_kernel(const u16* __restrict__ dO,
 #pragma unroll
     for (int ch = 0; ch < NCH; ++ch) {
       bf16x8_t a = read_rm_frag<C>(ldsDO, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
-      dp = mfma_32x32x16_bf16(a, vf[ch], dp);
+      bf16x8_t vf = *(const bf16x8_t*)(vrow + 16 * ch + 8 * (lane >> 5));
+      dp = mfma_32x32x16_bf16(a, vf, dp);
     }
     // dS = P * (dP - delta[q]) * scale  (keep layout)
     float ds[16];
