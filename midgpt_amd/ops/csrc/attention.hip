@@ -237,20 +237,16 @@ __global__ void attn_delta_kernel(const u16* __restrict__ dO,
 }
 
 // ===========================================================================
-// Backward: one WG = 128 k rows (32/wave); iterates q tiles >= diagonal.
-// dK/dV in registers; dQ via fp32 atomics. S recomputed from Q,K + LSE.
+// Backward, kernel A (dK/dV): one WG = 128 k rows (32/wave); iterates q
+// tiles >= its diagonal. dK/dV accumulate in registers (no atomics).
+// S and dS are recomputed from Q,K,LSE (standard flash recompute).
 // ===========================================================================
 template <int C>
-__global__ __launch_bounds__(256, 1) void attn_bwd_kernel(const u16* __restrict__ dO,
-                                const u16* __restrict__ q,
-                                const u16* __restrict__ k,
-                                const u16* __restrict__ v,
-                                const float* __restrict__ lse,
-                                const float* __restrict__ delta,
-                                float* __restrict__ dq32,
-                                u16* __restrict__ dk,
-                                u16* __restrict__ dv,
-                                int B, int H, int T) {
+__global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
+    const u16* __restrict__ dO, const u16* __restrict__ q,
+    const u16* __restrict__ k, const u16* __restrict__ v,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    u16* __restrict__ dk, u16* __restrict__ dv, int B, int H, int T) {
   constexpr int NCB = C / 32;
   constexpr int NCH = C / 16;
   const float scale = rsqrtf((float)C);
@@ -259,55 +255,29 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(const u16* __restrict_
   const long bh = blockIdx.x / nKB;
   const int lane = lane_id();
   const int w = wave_id();
-  const int kw0 = kb * 128 + 32 * w;    // wave's first k row
-  const int myk = kw0 + (lane & 31);    // lane's k row
+  const int kw0 = kb * 128 + 32 * w;
+  const int myk = kw0 + (lane & 31);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // [Qrm 32*C | Qt C*32 | dOrm 32*C | dOt C*32 | Kt 4*(C*32) | dS 4*(32*32)
-  //  | lse 32 | delta 32 ]  (bf16 tiles; dS bf16)
-  u16* ldsQ = (u16*)smem;
-  u16* ldsQt = ldsQ + 32 * C;
-  u16* ldsDO = ldsQt + C * 32;
-  u16* ldsDOt = ldsDO + 32 * C;
-  u16* ldsKt = ldsDOt + C * 32;          // per wave: + w*C*32
-  u16* ldsDS = ldsKt + 4 * C * 32;       // per wave: + w*32*32
-  float* ldsLse = (float*)(ldsDS + 4 * 32 * 32);
+  u16* ldsQ = (u16*)smem;              // [32*C] row-major
+  u16* ldsQt = ldsQ + 32 * C;          // [C*32] transposed
+  u16* ldsDO = ldsQt + C * 32;         // [32*C]
+  u16* ldsDOt = ldsDO + 32 * C;        // [C*32]
+  float* ldsLse = (float*)(ldsDOt + C * 32);
   float* ldsDelta = ldsLse + 32;
 
   const u16* qg = q + (bh * T) * C;
   const u16* kg = k + (bh * T) * C;
   const u16* vg = v + (bh * T) * C;
   const u16* dog = dO + (bh * T) * C;
-
-  // wave-owned K row fragments (B-fragment shape: contiguous 8 c).
-  // V fragments are re-read from global per q-tile (L2-resident) to keep
-  // register pressure under the 256-VGPR spill cliff.
-  bf16x8_t kf[NCH];
+  const u16* krow = kg + (long)myk * C;
   const u16* vrow = vg + (long)myk * C;
-  {
-    const u16* krow = kg + (long)myk * C;
-#pragma unroll
-    for (int ch = 0; ch < NCH; ++ch)
-      kf[ch] = *(const bf16x8_t*)(krow + 16 * ch + 8 * (lane >> 5));
-  }
-  // wave-private transposed K image for the dQ mfma (B operand needs
-  // column access): Kt[c][k_local 0..31]
-  {
-    u16* kt = ldsKt + w * C * 32;
-    for (int idx = lane * 8; idx < 32 * C; idx += WAVE * 8) {
-      const int row = idx / C, col = idx % C;
-      u16x8 val = *(const u16x8*)(kg + (long)(kw0 + row) * C + col);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *(u16*)((char*)(kt + (col + j) * 32) + swz_tr(col + j, row * 2)) = val[j];
-    }
-  }
 
   f32x16 dvacc[NCB], dkacc[NCB];
 #pragma unroll
   for (int cb = 0; cb < NCB; ++cb) { dvacc[cb] = (f32x16)(0.f); dkacc[cb] = (f32x16)(0.f); }
 
-  const int qt0 = kb * 128 / 32;  // first q tile (diagonal)
+  const int qt0 = kb * 4;  // diagonal q tile (kb*128/32)
   const int nqt = T / 32;
   for (int qt = qt0; qt < nqt; ++qt) {
     const int qbase = qt * 32;
@@ -323,14 +293,14 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(const u16* __restrict_
     __syncthreads();
     if (qbase + 31 < kw0) continue;  // fully masked for this wave
 
-    // S = Q x K^T  (D rows = q regs, cols = k lanes)
+    // S = Q x K^T (D rows = q regs, cols = k lanes); K frags from global
     f32x16 s = (f32x16)(0.f);
 #pragma unroll
     for (int ch = 0; ch < NCH; ++ch) {
       bf16x8_t a = read_rm_frag<C>(ldsQ, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
-      s = mfma_32x32x16_bf16(a, kf[ch], s);
+      bf16x8_t kf = *(const bf16x8_t*)(krow + 16 * ch + 8 * (lane >> 5));
+      s = mfma_32x32x16_bf16(a, kf, s);
     }
-    // P = exp(s*scale - lse[q]), masked
     float p[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -338,7 +308,7 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(const u16* __restrict_
       p[r] = (myk > qrow) ? 0.f
            : __expf(s[r] * scale - ldsLse[mfma_d_row(lane, r)]);
     }
-    // dV += P^T x dO   (A = packed P: rows k, kdim q)
+    // dV += P^T x dO
     bf16x8_t pf0 = dlayout_to_afrag(p);
     bf16x8_t pf1 = dlayout_to_afrag(p + 8);
 #pragma unroll
@@ -348,7 +318,7 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(const u16* __restrict_
       dvacc[cb] = mfma_32x32x16_bf16(pf0, b0, dvacc[cb]);
       dvacc[cb] = mfma_32x32x16_bf16(pf1, b1, dvacc[cb]);
     }
-    // dP = dO x V^T  (D rows = q regs, cols = k lanes)
+    // dP = dO x V^T; V frags from global (L2-resident)
     f32x16 dp = (f32x16)(0.f);
 #pragma unroll
     for (int ch = 0; ch < NCH; ++ch) {
@@ -356,12 +326,11 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(const u16* __restrict_
       bf16x8_t vf = *(const bf16x8_t*)(vrow + 16 * ch + 8 * (lane >> 5));
       dp = mfma_32x32x16_bf16(a, vf, dp);
     }
-    // dS = P * (dP - delta[q]) * scale  (keep layout)
+    // dS = P * (dP - delta[q]) * scale; dK += dS^T x Q
     float ds[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r)
       ds[r] = p[r] * (dp[r] - ldsDelta[mfma_d_row(lane, r)]) * scale;
-    // dK += dS^T x Q  (A = packed dS: rows k, kdim q; B = Qt)
     bf16x8_t df0 = dlayout_to_afrag(ds);
     bf16x8_t df1 = dlayout_to_afrag(ds + 8);
 #pragma unroll
@@ -371,43 +340,9 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(const u16* __restrict_
       dkacc[cb] = mfma_32x32x16_bf16(df0, b0, dkacc[cb]);
       dkacc[cb] = mfma_32x32x16_bf16(df1, b1, dkacc[cb]);
     }
-    // dQ: transpose dS through per-wave LDS, then dQ += dS x K
-    {
-      u16* dsl = ldsDS + w * 32 * 32;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int row = mfma_d_row(lane, r);  // q-local
-        *(u16*)((char*)(dsl + row * 32) + swz_tr(row, (lane & 31) * 2)) = f2b(ds[r]);
-      }
-      __builtin_amdgcn_s_waitcnt(0);
-      f32x16 dqa[NCB];
-#pragma unroll
-      for (int cb = 0; cb < NCB; ++cb) dqa[cb] = (f32x16)(0.f);
-      const u16* kt = ldsKt + w * C * 32;
-#pragma unroll
-      for (int kc = 0; kc < 2; ++kc) {
-        // A[q = lane&31][k = 16*kc + 8*(lane>>5) + j]
-        bf16x8_t a = *(const bf16x8_t*)((const char*)(dsl + (lane & 31) * 32) +
-                               swz_tr(lane & 31, (16 * kc + 8 * (lane >> 5)) * 2));
-#pragma unroll
-        for (int cb = 0; cb < NCB; ++cb) {
-          bf16x8_t b = read_tr_frag(kt, 32 * cb + (lane & 31),
-                                    (16 * kc + 8 * (lane >> 5)) * 2);
-          dqa[cb] = mfma_32x32x16_bf16(a, b, dqa[cb]);
-        }
-      }
-#pragma unroll
-      for (int cb = 0; cb < NCB; ++cb)
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int qrow = qbase + mfma_d_row(lane, r);
-          atomicAdd(dq32 + ((bh * T + qrow) * C + 32 * cb + (lane & 31)),
-                    dqa[cb][r]);
-        }
-    }
   }
 
-  // epilogue: dK, dV regs -> LDS bounce -> wide bf16 stores
+  // epilogue: LDS bounce -> wide bf16 stores (reuses the staging region)
   __syncthreads();
   float* ob = (float*)smem + w * 32 * 32;
   u16* dkg = dk + (bh * T + kw0) * C;
@@ -442,5 +377,145 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(const u16* __restrict_
       *(u16x8*)(out + (long)row * C + 32 * cb + c16 + 8) = o1;
       __builtin_amdgcn_s_waitcnt(0);
     }
+  }
+}
+
+// ===========================================================================
+// Backward, kernel B (dQ): one WG = 128 q rows (32/wave); iterates k tiles
+// up to its diagonal (fwd-shaped traversal). dQ accumulates in registers,
+// direct bf16 store — no atomics.
+// ===========================================================================
+template <int C>
+__global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
+    const u16* __restrict__ dO, const u16* __restrict__ q,
+    const u16* __restrict__ k, const u16* __restrict__ v,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    u16* __restrict__ dq, int B, int H, int T) {
+  constexpr int NCB = C / 32;
+  constexpr int NCH = C / 16;
+  const float scale = rsqrtf((float)C);
+  const int nQB = T / 128;
+  const int qb = blockIdx.x % nQB;
+  const long bh = blockIdx.x / nQB;
+  const int q0 = qb * 128;
+  const int lane = lane_id();
+  const int w = wave_id();
+  const int qw0 = q0 + 32 * w;
+  const int myq = qw0 + (lane & 31);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* ldsK = (u16*)smem;          // [32*C] row-major
+  u16* ldsV = ldsK + 32 * C;       // [32*C] row-major
+  u16* ldsKt = ldsV + 32 * C;      // [C*32] transposed
+  u16* ldsDS = ldsKt + C * 32;     // per wave + w*32*32 (bf16)
+
+  const u16* qg = q + (bh * T) * C;
+  const u16* kg = k + (bh * T) * C;
+  const u16* vg = v + (bh * T) * C;
+  const u16* dog = dO + (bh * T) * C;
+
+  // per-lane Q A-fragments (row q = lane&31 within the wave tile); dO
+  // fragments are re-read per tile from global (L2-resident) to stay under
+  // the 256-VGPR spill cliff.
+  bf16x8_t qf[NCH];
+  const u16* dorow = dog + (long)myq * C;
+  {
+    const u16* qrow = qg + (long)myq * C;
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      qf[ch] = *(const bf16x8_t*)(qrow + 16 * ch + 8 * (lane >> 5));
+  }
+  const float mylse = lse[bh * T + myq];
+  const float mydelta = delta[bh * T + myq];
+
+  f32x16 dqacc[NCB];
+#pragma unroll
+  for (int cb = 0; cb < NCB; ++cb) dqacc[cb] = (f32x16)(0.f);
+
+  const int nkt = (q0 + 128) / 32;
+  for (int kt = 0; kt < nkt; ++kt) {
+    const int k0 = kt * 32;
+    __syncthreads();
+    stage_rm<C>(kg + (long)k0 * C, ldsK);
+    stage_rm<C>(vg + (long)k0 * C, ldsV);
+    stage_tr<C>(kg + (long)k0 * C, ldsKt);
+    __syncthreads();
+    if (k0 > qw0 + 31) continue;  // beyond this wave's diagonal
+
+    // S = Q x K^T and dP = dO x V^T in one pass.
+    // A = per-lane Q/dO fragments (A[q=lane&31][c]), B = K/V rows from LDS
+    // (B[c][k=lane&31]); D rows = q (reg-mapped via mfma_d_row), cols = k.
+    f32x16 s = (f32x16)(0.f);
+    f32x16 dp = (f32x16)(0.f);
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch) {
+      bf16x8_t kfrag = read_rm_frag<C>(ldsK, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+      s = mfma_32x32x16_bf16(qf[ch], kfrag, s);
+      bf16x8_t vfrag = read_rm_frag<C>(ldsV, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+      bf16x8_t dof = *(const bf16x8_t*)(dorow + 16 * ch + 8 * (lane >> 5));
+      dp = mfma_32x32x16_bf16(dof, vfrag, dp);
+    }
+    // rows q are reg-mapped; cols k = lane&31. lse/delta per q row via shfl.
+    float ds[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = q0 + 32 * w + mfma_d_row(lane, r);
+      const float l_r = shfl32(mylse, mfma_d_row(lane, r));
+      const float d_r = shfl32(mydelta, mfma_d_row(lane, r));
+      const int kcol = k0 + (lane & 31);
+      float pv = (kcol > qrow) ? 0.f : __expf(s[r] * scale - l_r);
+      ds[r] = pv * (dp[r] - d_r) * scale;
+    }
+    // transpose dS through per-wave LDS -> A-frags A[q = lane&31][k]
+    u16* dsl = ldsDS + w * 32 * 32;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = mfma_d_row(lane, r);  // q-local
+      *(u16*)((char*)(dsl + row * 32) + swz_tr(row, (lane & 31) * 2)) = f2b(ds[r]);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      bf16x8_t a = *(const bf16x8_t*)((const char*)(dsl + (lane & 31) * 32) +
+                                      swz_tr(lane & 31, (16 * kc + 8 * (lane >> 5)) * 2));
+#pragma unroll
+      for (int cb = 0; cb < NCB; ++cb) {
+        bf16x8_t b = read_tr_frag(ldsKt, 32 * cb + (lane & 31),
+                                  (16 * kc + 8 * (lane >> 5)) * 2);
+        dqacc[cb] = mfma_32x32x16_bf16(a, b, dqacc[cb]);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);  // dsl reads done before next overwrite
+  }
+
+  // epilogue: LDS bounce -> wide stores
+  __syncthreads();
+  float* ob = (float*)smem + w * 32 * 32;
+  u16* dqg = dq + (bh * T + qw0) * C;
+#pragma unroll
+  for (int cb = 0; cb < NCB; ++cb) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = mfma_d_row(lane, r);
+      *(float*)((char*)ob + row * 128 + (((lane & 31) * 4) ^ ((row & 7) << 4))) =
+          dqacc[cb][r];
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    const int row = lane & 31;
+    const int c16 = 16 * (lane >> 5);
+    float tmp[16];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      f32x4 t = *(const f32x4*)((char*)ob + row * 128 +
+                                (((c16 + 4 * i) * 4) ^ ((row & 7) << 4)));
+      tmp[4 * i] = t[0]; tmp[4 * i + 1] = t[1];
+      tmp[4 * i + 2] = t[2]; tmp[4 * i + 3] = t[3];
+    }
+    u16x8 o0, o1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { o0[j] = f2b(tmp[j]); o1[j] = f2b(tmp[8 + j]); }
+    *(u16x8*)(dqg + (long)row * C + 32 * cb + c16) = o0;
+    *(u16x8*)(dqg + (long)row * C + 32 * cb + c16 + 8) = o1;
+    __builtin_amdgcn_s_waitcnt(0);
   }
 }

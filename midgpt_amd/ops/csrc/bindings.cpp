@@ -220,28 +220,29 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
                      (const u16*)dO.data_ptr(), (const u16*)o.data_ptr(),
                      delta.data_ptr<float>(), N, C);
   launch_check();
-  auto dq32 = torch::zeros({B, H, T, C}, q.options().dtype(torch::kFloat));
+  auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
   long grid = (long)B * H * (T / 128);
-  size_t smem = (size_t)(2 * 32 * C + 2 * C * 32 + 4 * C * 32 + 4 * 32 * 32) * 2 + 64 * 4;
-  if (C == 128) {
-    hipLaunchKernelGGL((attn_bwd_kernel<128>), dim3(grid), dim3(256), smem, cur_stream(),
-                       (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),
-                       (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       dq32.data_ptr<float>(), (u16*)dk.data_ptr(),
-                       (u16*)dv.data_ptr(), B, H, T);
-  } else {
-    hipLaunchKernelGGL((attn_bwd_kernel<64>), dim3(grid), dim3(256), smem, cur_stream(),
-                       (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),
-                       (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       dq32.data_ptr<float>(), (u16*)dk.data_ptr(),
-                       (u16*)dv.data_ptr(), B, H, T);
-  }
+  size_t smem_a = (size_t)(4 * 32 * C) * 2 + 64 * 4;  // Qrm+Qt+dOrm+dOt+lse/delta
+  size_t smem_b = (size_t)(2 * 32 * C + C * 32 + 4 * 32 * 32) * 2;  // K,V,Kt,dS
+#define LAUNCH_BWD(CC)                                                          \
+  do {                                                                          \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC>), dim3(grid), dim3(256), smem_a,\
+                       cur_stream(), (const u16*)dO.data_ptr(),                 \
+                       (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),      \
+                       (const u16*)v.data_ptr(), lse.data_ptr<float>(),         \
+                       delta.data_ptr<float>(), (u16*)dk.data_ptr(),            \
+                       (u16*)dv.data_ptr(), B, H, T);                           \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<CC>), dim3(grid), dim3(256), smem_b, \
+                       cur_stream(), (const u16*)dO.data_ptr(),                 \
+                       (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),      \
+                       (const u16*)v.data_ptr(), lse.data_ptr<float>(),         \
+                       delta.data_ptr<float>(), (u16*)dq.data_ptr(), B, H, T);  \
+  } while (0)
+  if (C == 128) LAUNCH_BWD(128); else LAUNCH_BWD(64);
+#undef LAUNCH_BWD
   launch_check();
-  auto dq = dq32.to(torch::kBFloat16);
   return {dq, dk, dv};
 }
 
