@@ -1,5 +1,7 @@
 """Microbenchmarks for the HIP hot kernels (run on the GPU box).
 Prints per-kernel time and effective TF/s or TB/s."""
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import argparse
 import time
 
