@@ -74,7 +74,7 @@ std::vector<torch::Tensor> qkv_prep_fwd(torch::Tensor qkv, torch::Tensor qw,
   CHECK_GPU(qkv);
   TORCH_CHECK(qkv.scalar_type() == torch::kBFloat16, "qkv must be bf16");
   int B = qkv.size(0), T = qkv.size(1), H = qkv.size(3), C = qkv.size(4);
-  TORCH_CHECK(C % 2 == 0 && C <= 128, "head dim must be even and <= 128");
+  TORCH_CHECK(C % 8 == 0 && C <= 128, "head dim must be %8==0 and <= 128");
   auto opt = qkv.options();
   auto q = torch::empty({B, H, T, C}, opt);
   auto k = torch::empty({B, H, T, C}, opt);

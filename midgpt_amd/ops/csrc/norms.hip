@@ -181,8 +181,9 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict
 // Fused QK-LayerNorm + RoPE over packed QKV (K3+K4).
 // qkv: (B, T, 3, H, C) bf16 -> q,k (B,H,T,C) LN'd + RoPE'd, v transposed.
 // stats: (B,H,T,2) fp32 {mean, invstd} for q and k.
-// One wave per (b,t,h,role) row; lane i < C/2 owns the interleaved RoPE
-// pair (2i, 2i+1) so the rotation is lane-local.
+// Vectorized: C/8 lanes per row, u16x8 loads (8 elems = 4 interleaved RoPE
+// pairs per lane, so the GPT-J rotation stays lane-local); a wave covers
+// 64/(C/8) rows; reductions via width-(C/8) shuffles.
 // ============================================================
 __global__ void qkv_prep_fwd_kernel(const u16* __restrict__ qkv,
                                     const float* __restrict__ qw,
@@ -194,56 +195,60 @@ __global__ void qkv_prep_fwd_kernel(const u16* __restrict__ qkv,
                                     float* __restrict__ qstats,
                                     float* __restrict__ kstats,
                                     int B, int T, int H, int C, float eps) {
-  const int lane = lane_id();
+  const int LPR = C / 8;                 // lanes per row (8 or 16)
+  const int RPW = WAVE / LPR;            // rows per wave
+  const int sub = lane_id() % LPR;       // lane's slot within its row
+  const int rsub = lane_id() / LPR;      // which row of the wave
   const long nrows = (long)B * T * H * 3;
-  const long row0 = (long)blockIdx.x * (blockDim.x / WAVE) + wave_id();
-  const long rstep = (long)gridDim.x * (blockDim.x / WAVE);
+  const long row0 = ((long)blockIdx.x * (blockDim.x / WAVE) + wave_id()) * RPW + rsub;
+  const long rstep = (long)gridDim.x * (blockDim.x / WAVE) * RPW;
   for (long row = row0; row < nrows; row += rstep) {
-    // row index order: (b, t, role, h) — matches qkv memory layout
     const int h = row % H;
     const int role = (row / H) % 3;
     const long t = (row / ((long)3 * H)) % T;
     const long b = row / ((long)3 * H * T);
-    const u16* src = qkv + ((((b * T + t) * 3 + role) * H + h) * C);
-    // output (B,H,T,C)
-    const long out_off = (((b * H + h) * T + t) * C);
-    if (role == 2) {  // V: straight transpose copy, 2 bf16/lane (C<=128)
-      for (int i = lane * 2; i + 1 < C; i += WAVE * 2) {
-        *(u16x2*)(v + out_off + i) = *(const u16x2*)(src + i);
-      }
-      if (C & 1) { if (lane == 0) v[out_off + C - 1] = src[C - 1]; }
+    const u16* src = qkv + ((((b * T + t) * 3 + role) * H + h) * C) + sub * 8;
+    const long out_off = (((b * H + h) * T + t) * C) + sub * 8;
+    u16x8 raw = *(const u16x8*)src;
+    if (role == 2) {  // V: straight transpose copy
+      *(u16x8*)(v + out_off) = raw;
       continue;
     }
-    // Q or K: LayerNorm over C (weight, no bias) then RoPE.
     const float* w = role == 0 ? qw : kw;
     u16* dst = role == 0 ? q : k;
     float* stats = role == 0 ? qstats : kstats;
-    const int P = C / 2;  // pairs
-    float x0 = 0.f, x1 = 0.f;
-    if (lane < P) {
-      x0 = b2f(src[2 * lane]);
-      x1 = b2f(src[2 * lane + 1]);
-    }
-    float s = group_sum<WAVE>(x0 + x1);
-    float mu = s / C;
-    float d0 = lane < P ? x0 - mu : 0.f, d1 = lane < P ? x1 - mu : 0.f;
-    float ss = group_sum<WAVE>(d0 * d0 + d1 * d1);
-    float invstd = rsqrtf(ss / C + eps);
-    if (lane == 0) {
+    float x[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) x[j] = b2f(raw[j]);
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s += x[j];
+    // width-LPR reduction within the row's lane group
+    for (int o = LPR / 2; o > 0; o >>= 1) s += __shfl_xor(s, o, LPR);
+    const float mu = s / C;
+    float ss = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { x[j] -= mu; ss += x[j] * x[j]; }
+    for (int o = LPR / 2; o > 0; o >>= 1) ss += __shfl_xor(ss, o, LPR);
+    const float invstd = rsqrtf(ss / C + eps);
+    if (sub == 0) {
       stats[2 * ((b * H + h) * T + t)] = mu;
       stats[2 * ((b * H + h) * T + t) + 1] = invstd;
     }
-    if (lane < P) {
-      float n0 = d0 * invstd * w[2 * lane];
-      float n1 = d1 * invstd * w[2 * lane + 1];
-      float sn = sin_t[t * P + lane];
-      float cs = cos_t[t * P + lane];
-      // out = x*cos + rotate_every_two(x)*sin ; rot([a,b]) = [-b, a]
-      float o0 = n0 * cs - n1 * sn;
-      float o1 = n1 * cs + n0 * sn;
-      u16x2 o; o.x = f2b(o0); o.y = f2b(o1);
-      *(u16x2*)(dst + out_off + 2 * lane) = o;
+    // normalize + weight + RoPE (pairs p = 4*sub + 0..3)
+    const float* srow = sin_t + t * (C / 2) + sub * 4;
+    const float* crow = cos_t + t * (C / 2) + sub * 4;
+    f32x4 sn = *(const f32x4*)srow;
+    f32x4 cs = *(const f32x4*)crow;
+    u16x8 out;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      float n0 = x[2 * p] * invstd * w[sub * 8 + 2 * p];
+      float n1 = x[2 * p + 1] * invstd * w[sub * 8 + 2 * p + 1];
+      out[2 * p] = f2b(n0 * cs[p] - n1 * sn[p]);
+      out[2 * p + 1] = f2b(n1 * cs[p] + n0 * sn[p]);
     }
+    *(u16x8*)(dst + out_off) = out;
   }
 }
 
@@ -262,10 +267,13 @@ __global__ void qkv_prep_bwd_kernel(const u16* __restrict__ dq,
                                     float* __restrict__ dqw_partial,
                                     float* __restrict__ dkw_partial,
                                     int B, int T, int H, int C) {
-  const int lane = lane_id();
+  const int LPR = C / 8;
+  const int RPW = WAVE / LPR;
+  const int sub = lane_id() % LPR;
+  const int rsub = lane_id() / LPR;
   const long nrows = (long)B * T * H * 3;
-  const long row0 = (long)blockIdx.x * (blockDim.x / WAVE) + wave_id();
-  const long rstep = (long)gridDim.x * (blockDim.x / WAVE);
+  const long row0 = ((long)blockIdx.x * (blockDim.x / WAVE) + wave_id()) * RPW + rsub;
+  const long rstep = (long)gridDim.x * (blockDim.x / WAVE) * RPW;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* smem = (float*)smem_raw;  // [2][C] partial dqw / dkw
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) smem[i] = 0.f;
@@ -275,45 +283,52 @@ __global__ void qkv_prep_bwd_kernel(const u16* __restrict__ dq,
     const int role = (row / H) % 3;
     const long t = (row / ((long)3 * H)) % T;
     const long b = row / ((long)3 * H * T);
-    u16* dst = dqkv + ((((b * T + t) * 3 + role) * H + h) * C);
-    const long in_off = (((b * H + h) * T + t) * C);
+    u16* dst = dqkv + ((((b * T + t) * 3 + role) * H + h) * C) + sub * 8;
+    const long in_off = (((b * H + h) * T + t) * C) + sub * 8;
     if (role == 2) {
-      for (int i = lane * 2; i + 1 < C; i += WAVE * 2)
-        *(u16x2*)(dst + i) = *(const u16x2*)(dv + in_off + i);
+      *(u16x8*)dst = *(const u16x8*)(dv + in_off);
       continue;
     }
     const float* w = role == 0 ? qw : kw;
     const u16* dyp = role == 0 ? dq : dk;
     const float* stats = role == 0 ? qstats : kstats;
     float* dwp = role == 0 ? smem : smem + C;
-    const int P = C / 2;
-    float g0 = 0.f, g1 = 0.f, xh0 = 0.f, xh1 = 0.f, dn0 = 0.f, dn1 = 0.f;
     const float mu = stats[2 * ((b * H + h) * T + t)];
     const float invstd = stats[2 * ((b * H + h) * T + t) + 1];
-    if (lane < P) {
+    u16x8 rawdy = *(const u16x8*)(dyp + in_off);
+    u16x8 rawx = *(const u16x8*)(qkv + ((((b * T + t) * 3 + role) * H + h) * C) + sub * 8);
+    const float* srow = sin_t + t * (C / 2) + sub * 4;
+    const float* crow = cos_t + t * (C / 2) + sub * 4;
+    f32x4 sn = *(const f32x4*)srow;
+    f32x4 cs = *(const f32x4*)crow;
+    float dn[8], xh[8], g[8];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
       // inverse RoPE rotation: dn = dy*cos + rot^T(dy*sin)
-      float d0 = b2f(dyp[in_off + 2 * lane]);
-      float d1 = b2f(dyp[in_off + 2 * lane + 1]);
-      float sn = sin_t[t * P + lane];
-      float cs = cos_t[t * P + lane];
-      dn0 = d0 * cs + d1 * sn;
-      dn1 = d1 * cs - d0 * sn;
-      const u16* src = qkv + ((((b * T + t) * 3 + role) * H + h) * C);
-      xh0 = (b2f(src[2 * lane]) - mu) * invstd;
-      xh1 = (b2f(src[2 * lane + 1]) - mu) * invstd;
-      g0 = dn0 * w[2 * lane];
-      g1 = dn1 * w[2 * lane + 1];
+      float d0 = b2f(rawdy[2 * p]), d1 = b2f(rawdy[2 * p + 1]);
+      dn[2 * p] = d0 * cs[p] + d1 * sn[p];
+      dn[2 * p + 1] = d1 * cs[p] - d0 * sn[p];
     }
-    float mean_g = group_sum<WAVE>(g0 + g1) / C;
-    float mean_gx = group_sum<WAVE>(g0 * xh0 + g1 * xh1) / C;
-    if (lane < P) {
-      float dx0 = invstd * (g0 - mean_g - xh0 * mean_gx);
-      float dx1 = invstd * (g1 - mean_g - xh1 * mean_gx);
-      u16x2 o; o.x = f2b(dx0); o.y = f2b(dx1);
-      *(u16x2*)(dst + 2 * lane) = o;
-      atomicAdd(&dwp[2 * lane], dn0 * xh0);
-      atomicAdd(&dwp[2 * lane + 1], dn1 * xh1);
+    float sum_g = 0.f, sum_gx = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      xh[j] = (b2f(rawx[j]) - mu) * invstd;
+      g[j] = dn[j] * w[sub * 8 + j];
+      sum_g += g[j];
+      sum_gx += g[j] * xh[j];
     }
+    for (int o = LPR / 2; o > 0; o >>= 1) {
+      sum_g += __shfl_xor(sum_g, o, LPR);
+      sum_gx += __shfl_xor(sum_gx, o, LPR);
+    }
+    const float mean_g = sum_g / C, mean_gx = sum_gx / C;
+    u16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      out[j] = f2b(invstd * (g[j] - mean_g - xh[j] * mean_gx));
+      atomicAdd(&dwp[sub * 8 + j], dn[j] * xh[j]);
+    }
+    *(u16x8*)dst = out;
   }
   __syncthreads();
   for (int i = threadIdx.x; i < C; i += blockDim.x) {
