@@ -36,20 +36,20 @@ DEVINL int swz_tr(int row, int byte_in_row) {
 // Cooperative staging helpers (256 threads)
 // ---------------------------------------------------------------------------
 // Row-major 32 x C tile from global (row stride C) into swizzled LDS.
-template <int C>
+template <int C, int NT>
 DEVINL void stage_rm(const u16* __restrict__ g, u16* lds) {
 #pragma unroll
-  for (int idx = threadIdx.x * 8; idx < 32 * C; idx += 256 * 8) {
+  for (int idx = threadIdx.x * 8; idx < 32 * C; idx += NT * 8) {
     const int row = idx / C, col = idx % C;
     u16x8 val = *(const u16x8*)(g + row * C + col);
     *(u16x8*)((char*)lds + row * C * 2 + swz_rm<C>(row, col * 2)) = val;
   }
 }
 // Transposed: global 32 x C (row stride C) -> LDS [C][32] swizzled.
-template <int C>
+template <int C, int NT>
 DEVINL void stage_tr(const u16* __restrict__ g, u16* lds) {
 #pragma unroll
-  for (int idx = threadIdx.x * 8; idx < 32 * C; idx += 256 * 8) {
+  for (int idx = threadIdx.x * 8; idx < 32 * C; idx += NT * 8) {
     const int row = idx / C, col = idx % C;  // row = kv/q index, col = c
     u16x8 val = *(const u16x8*)(g + row * C + col);
 #pragma unroll
@@ -75,8 +75,8 @@ DEVINL float shfl32(float v, int src) { return __shfl(v, src, 32); }
 // ===========================================================================
 // Forward
 // ===========================================================================
-template <int C>
-__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(const u16* __restrict__ q,
+template <int C, int NW>
+__global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restrict__ q,
                                 const u16* __restrict__ k,
                                 const u16* __restrict__ v,
                                 u16* __restrict__ o, float* __restrict__ lse,
@@ -84,10 +84,11 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(const u16* __restrict_
   constexpr int NCB = C / 32;   // 32-col c-blocks
   constexpr int NCH = C / 16;   // 16-deep mfma chunks
   const float scale = rsqrtf((float)C);
-  const int nQB = T / 128;
-  const int qb = blockIdx.x % nQB;
-  const long bh = blockIdx.x / nQB;  // b*H + h
-  const int q0 = qb * 128;
+  // qb-outermost grid order: all q-blocks of one (b,h) land on the same
+  // XCD (b%8 dispatch) for K/V L2 reuse (guide T1).
+  const long bh = blockIdx.x % ((long)B * H);
+  const int qb = blockIdx.x / (B * H);
+  const int q0 = qb * (NW * 32);
   const int lane = lane_id();
   const int w = wave_id();
   const int qw0 = q0 + 32 * w;          // this wave's first q row
@@ -116,16 +117,16 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(const u16* __restrict_
   for (int cb = 0; cb < NCB; ++cb) oacc[cb] = (f32x16)(0.f);
   float m = -1e30f, lsum = 0.f;
 
-  const int nkt = (q0 + 128) / 32;
-  stage_rm<C>(kg, ldsK);
-  stage_tr<C>(vg, ldsVt);
+  const int nkt = (q0 + NW * 32) / 32;
+  stage_rm<C, NW * 64>(kg, ldsK);
+  stage_tr<C, NW * 64>(vg, ldsVt);
   __syncthreads();
 
   for (int kt = 0; kt < nkt; ++kt) {
     const int buf = kt & 1;
     if (kt + 1 < nkt) {
-      stage_rm<C>(kg + (long)(kt + 1) * 32 * C, ldsK + (1 - buf) * 32 * C);
-      stage_tr<C>(vg + (long)(kt + 1) * 32 * C, ldsVt + (1 - buf) * C * 32);
+      stage_rm<C, NW * 64>(kg + (long)(kt + 1) * 32 * C, ldsK + (1 - buf) * 32 * C);
+      stage_tr<C, NW * 64>(vg + (long)(kt + 1) * 32 * C, ldsVt + (1 - buf) * C * 32);
     }
     const int k0 = kt * 32;
     if (k0 <= qw0 + 31) {  // wave-uniform: tile not fully masked for this wave
@@ -241,8 +242,8 @@ __global__ void attn_delta_kernel(const u16* __restrict__ dO,
 // tiles >= its diagonal. dK/dV accumulate in registers (no atomics).
 // S and dS are recomputed from Q,K,LSE (standard flash recompute).
 // ===========================================================================
-template <int C>
-__global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
+template <int C, int NW>
+__global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
     const u16* __restrict__ dO, const u16* __restrict__ q,
     const u16* __restrict__ k, const u16* __restrict__ v,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -250,12 +251,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   constexpr int NCB = C / 32;
   constexpr int NCH = C / 16;
   const float scale = rsqrtf((float)C);
-  const int nKB = T / 128;
-  const int kb = blockIdx.x % nKB;
-  const long bh = blockIdx.x / nKB;
+  const long bh = blockIdx.x % ((long)B * H);
+  const int kb = blockIdx.x / (B * H);
   const int lane = lane_id();
   const int w = wave_id();
-  const int kw0 = kb * 128 + 32 * w;
+  const int kw0 = kb * (NW * 32) + 32 * w;
   const int myk = kw0 + (lane & 31);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -277,15 +277,15 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
   for (int cb = 0; cb < NCB; ++cb) { dvacc[cb] = (f32x16)(0.f); dkacc[cb] = (f32x16)(0.f); }
 
-  const int qt0 = kb * 4;  // diagonal q tile (kb*128/32)
+  const int qt0 = kb * NW;  // diagonal q tile
   const int nqt = T / 32;
   for (int qt = qt0; qt < nqt; ++qt) {
     const int qbase = qt * 32;
     __syncthreads();
-    stage_rm<C>(qg + (long)qbase * C, ldsQ);
-    stage_tr<C>(qg + (long)qbase * C, ldsQt);
-    stage_rm<C>(dog + (long)qbase * C, ldsDO);
-    stage_tr<C>(dog + (long)qbase * C, ldsDOt);
+    stage_rm<C, NW * 64>(qg + (long)qbase * C, ldsQ);
+    stage_tr<C, NW * 64>(qg + (long)qbase * C, ldsQt);
+    stage_rm<C, NW * 64>(dog + (long)qbase * C, ldsDO);
+    stage_tr<C, NW * 64>(dog + (long)qbase * C, ldsDOt);
     if (threadIdx.x < 32) {
       ldsLse[threadIdx.x] = lse[bh * T + qbase + threadIdx.x];
       ldsDelta[threadIdx.x] = delta[bh * T + qbase + threadIdx.x];
@@ -385,8 +385,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 // up to its diagonal (fwd-shaped traversal). dQ accumulates in registers,
 // direct bf16 store — no atomics.
 // ===========================================================================
-template <int C>
-__global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
+template <int C, int NW>
+__global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
     const u16* __restrict__ dO, const u16* __restrict__ q,
     const u16* __restrict__ k, const u16* __restrict__ v,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -394,10 +394,9 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   constexpr int NCB = C / 32;
   constexpr int NCH = C / 16;
   const float scale = rsqrtf((float)C);
-  const int nQB = T / 128;
-  const int qb = blockIdx.x % nQB;
-  const long bh = blockIdx.x / nQB;
-  const int q0 = qb * 128;
+  const long bh = blockIdx.x % ((long)B * H);
+  const int qb = blockIdx.x / (B * H);
+  const int q0 = qb * (NW * 32);
   const int lane = lane_id();
   const int w = wave_id();
   const int qw0 = q0 + 32 * w;
@@ -432,13 +431,13 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 #pragma unroll
   for (int cb = 0; cb < NCB; ++cb) dqacc[cb] = (f32x16)(0.f);
 
-  const int nkt = (q0 + 128) / 32;
+  const int nkt = (q0 + NW * 32) / 32;
   for (int kt = 0; kt < nkt; ++kt) {
     const int k0 = kt * 32;
     __syncthreads();
-    stage_rm<C>(kg + (long)k0 * C, ldsK);
-    stage_rm<C>(vg + (long)k0 * C, ldsV);
-    stage_tr<C>(kg + (long)k0 * C, ldsKt);
+    stage_rm<C, NW * 64>(kg + (long)k0 * C, ldsK);
+    stage_rm<C, NW * 64>(vg + (long)k0 * C, ldsV);
+    stage_tr<C, NW * 64>(kg + (long)k0 * C, ldsKt);
     __syncthreads();
     if (k0 > qw0 + 31) continue;  // beyond this wave's diagonal
 

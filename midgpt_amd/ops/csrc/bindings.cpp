@@ -192,19 +192,21 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   TORCH_CHECK(C == 64 || C == 128, "attn: head dim 64 or 128 (got ", C, ")");
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, T}, q.options().dtype(torch::kFloat));
-  long grid = (long)B * H * (T / 128);
-  size_t smem = std::max((size_t)(4 * 32 * C * 2), (size_t)(4 * 32 * 32 * 4));
-  if (C == 128) {
-    hipLaunchKernelGGL((attn_fwd_kernel<128>), dim3(grid), dim3(256), smem, cur_stream(),
-                       (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),
-                       (const u16*)v.data_ptr(), (u16*)o.data_ptr(),
-                       lse.data_ptr<float>(), B, H, T);
-  } else {
-    hipLaunchKernelGGL((attn_fwd_kernel<64>), dim3(grid), dim3(256), smem, cur_stream(),
-                       (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),
-                       (const u16*)v.data_ptr(), (u16*)o.data_ptr(),
-                       lse.data_ptr<float>(), B, H, T);
-  }
+  // 8-wave WGs (256 q rows) halve K/V re-read bandwidth vs 4-wave; fall
+  // back to 4 waves when T % 256 != 0.
+  const int NW = (T % 256 == 0) ? 8 : 4;
+  long grid = (long)B * H * (T / (NW * 32));
+  size_t smem = std::max((size_t)(4 * 32 * C * 2), (size_t)(NW * 32 * 32 * 4));
+#define LAUNCH_FWD(CC, NN)                                                      \
+  hipLaunchKernelGGL((attn_fwd_kernel<CC, NN>), dim3(grid), dim3(NN * 64), smem,\
+                     cur_stream(), (const u16*)q.data_ptr(),                    \
+                     (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),        \
+                     (u16*)o.data_ptr(), lse.data_ptr<float>(), B, H, T)
+  if (C == 128 && NW == 8) LAUNCH_FWD(128, 8);
+  else if (C == 128) LAUNCH_FWD(128, 4);
+  else if (NW == 8) LAUNCH_FWD(64, 8);
+  else LAUNCH_FWD(64, 4);
+#undef LAUNCH_FWD
   launch_check();
   return {o, lse};
 }
@@ -223,24 +225,30 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
-  long grid = (long)B * H * (T / 128);
-  size_t smem_a = (size_t)(4 * 32 * C) * 2 + 64 * 4;  // Qrm+Qt+dOrm+dOt+lse/delta
-  size_t smem_b = (size_t)(2 * 32 * C + C * 32 + 4 * 32 * 32) * 2;  // K,V,Kt,dS
-#define LAUNCH_BWD(CC)                                                          \
+  const int NW = (T % 256 == 0) ? 8 : 4;
+  long grid = (long)B * H * (T / (NW * 32));
+  size_t smem_a = std::max((size_t)(4 * 32 * C) * 2 + 64 * 4,
+                           (size_t)(NW * 32 * 32 * 4));
+  size_t smem_b = std::max((size_t)(2 * 32 * C + C * 32) * 2 + NW * 32 * 32 * 2,
+                           (size_t)(NW * 32 * 32 * 4));
+#define LAUNCH_BWD(CC, NN)                                                      \
   do {                                                                          \
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC>), dim3(grid), dim3(256), smem_a,\
-                       cur_stream(), (const u16*)dO.data_ptr(),                 \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NN>), dim3(grid), dim3(NN * 64),\
+                       smem_a, cur_stream(), (const u16*)dO.data_ptr(),         \
                        (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),      \
                        (const u16*)v.data_ptr(), lse.data_ptr<float>(),         \
                        delta.data_ptr<float>(), (u16*)dk.data_ptr(),            \
                        (u16*)dv.data_ptr(), B, H, T);                           \
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<CC>), dim3(grid), dim3(256), smem_b, \
-                       cur_stream(), (const u16*)dO.data_ptr(),                 \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<CC, NN>), dim3(grid), dim3(NN * 64), \
+                       smem_b, cur_stream(), (const u16*)dO.data_ptr(),         \
                        (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),      \
                        (const u16*)v.data_ptr(), lse.data_ptr<float>(),         \
                        delta.data_ptr<float>(), (u16*)dq.data_ptr(), B, H, T);  \
   } while (0)
-  if (C == 128) LAUNCH_BWD(128); else LAUNCH_BWD(64);
+  if (C == 128 && NW == 8) LAUNCH_BWD(128, 8);
+  else if (C == 128) LAUNCH_BWD(128, 4);
+  else if (NW == 8) LAUNCH_BWD(64, 8);
+  else LAUNCH_BWD(64, 4);
 #undef LAUNCH_BWD
   launch_check();
   return {dq, dk, dv};
