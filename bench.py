@@ -34,7 +34,11 @@ def main():
     p.add_argument("--local-batch", type=int, default=128,
                    help="sequences per GPU per optimizer step")
     p.add_argument("--micro-batch", type=int, default=None,
-                   help="microbatch size (default: local-batch, i.e. G=1)")
+                   help="microbatch size (default: 64 at T<=1024, else 8)")
+    p.add_argument("--remat", action="store_true",
+                   help="per-block activation recompute (default OFF for the "
+                        "bench: 288 GB HBM fits stored activations at T=1024, "
+                        "saving the recompute forward)")
     args = p.parse_args()
 
     rank, world, device = pdist.init_distributed()
@@ -45,8 +49,10 @@ def main():
     config = load_config(args.config)
     config.synthetic_data = True
     config.rundir = ""
+    config.remat = args.remat or config.model_config.block_size > 1024
     mc = config.model_config
-    micro = args.micro_batch or args.local_batch
+    micro = args.micro_batch or min(args.local_batch,
+                                    64 if mc.block_size <= 1024 else 8)
     assert args.local_batch % micro == 0
     g_accum = args.local_batch // micro
     config.batch_size = micro * n
@@ -130,6 +136,7 @@ def main():
                 "seq_len": mc.block_size,
                 "parallelism": (f"zero-dp{n}" if engine.zero else f"dp{n}"),
                 "g_accum_iters": g_accum,
+                "remat": config.remat,
                 "mfu_vs_2.5pf_dense": round(mfu, 4),
                 "last_loss": round(float(loss), 4),
             },
