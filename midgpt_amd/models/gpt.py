@@ -20,7 +20,9 @@ natively batched (B, T, ...) so each op is ONE kernel launch on the GPU.
 """
 from __future__ import annotations
 
+import contextlib
 import math
+import os
 
 import torch
 import torch.nn as nn
@@ -29,6 +31,16 @@ import torch.nn.functional as F
 from midgpt_amd import ops
 from midgpt_amd.config import GPTConfig
 from midgpt_amd.ops.reference import rope_tables
+
+
+_SCOPES = os.environ.get("MIDGPT_SCOPES") == "1"
+
+
+def _scope(name):
+    """Named trace scopes (reference jax.named_scope parity); enabled with
+    MIDGPT_SCOPES=1 so the hot path stays annotation-free by default."""
+    return torch.profiler.record_function(name) if _SCOPES \
+        else contextlib.nullcontext()
 
 
 def _init_linear_(w: torch.Tensor, in_features: int, generator=None):
@@ -67,6 +79,10 @@ class CausalSelfAttention(nn.Module):
         self.resid_dropout = nn.Dropout(config.dropout)
 
     def forward(self, x, sin, cos):
+        with _scope("causal_sa"):
+            return self._forward(x, sin, cos)
+
+    def _forward(self, x, sin, cos):
         B, T, D = x.shape
         qkv = self.c_attn(x).view(B, T, 3, self.n_head, self.head_dim)
         q, k, v = ops.qkv_prep(qkv, self.q_ln_weight, self.k_ln_weight,
@@ -92,7 +108,9 @@ class MLP(nn.Module):
         self.dropout = nn.Dropout(config.dropout)
 
     def forward(self, x):
-        return self.dropout(self.c_proj(F.gelu(self.c_fc(x), approximate="tanh")))
+        with _scope("mlp"):
+            return self.dropout(
+                self.c_proj(F.gelu(self.c_fc(x), approximate="tanh")))
 
 
 class Block(nn.Module):
@@ -104,9 +122,10 @@ class Block(nn.Module):
         self.mlp = MLP(config, generator)
 
     def forward(self, x, sin, cos):
-        x = x + self.attn(ops.rmsnorm(x, None, 1e-6), sin, cos)
-        x = x + self.mlp(ops.rmsnorm(x, None, 1e-6))
-        return x
+        with _scope("block"):
+            x = x + self.attn(ops.rmsnorm(x, None, 1e-6), sin, cos)
+            x = x + self.mlp(ops.rmsnorm(x, None, 1e-6))
+            return x
 
 
 class GPT(nn.Module):

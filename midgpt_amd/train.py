@@ -106,15 +106,34 @@ def train(config: ExperimentConfig):
                               decay_steps=config.lr_decay_steps,
                               min_lr=config.min_lr)
         x, y = loader.batch("train", local_bs, config.g_accum_iters)
-        loss_sum = 0.0
-        for g in range(config.g_accum_iters):
-            xg, yg = x[g].to(device, non_blocking=True), y[g].to(device, non_blocking=True)
-            loss = model.loss(xg, yg)
-            loss.backward()
-            engine.microstep_end()
-            loss_sum += float(loss)
-        engine.step(lr, config.g_accum_iters)
-        loss_step = loss_sum / config.g_accum_iters
+
+        def one_step():
+            s = 0.0
+            for g in range(config.g_accum_iters):
+                xg = x[g].to(device, non_blocking=True)
+                yg = y[g].to(device, non_blocking=True)
+                loss = model.loss(xg, yg)
+                loss.backward()
+                engine.microstep_end()
+                s += float(loss)
+            engine.step(lr, config.g_accum_iters)
+            return s / config.g_accum_iters
+
+        if config.debug and it == first_step and config.rundir:
+            # reference parity: --debug traces step 0 (src/train.py:205-211);
+            # here a torch.profiler (rocprof-sdk-backed) chrome trace.
+            from torch.profiler import ProfilerActivity, profile
+            acts = [ProfilerActivity.CPU]
+            if device.type == "cuda":
+                acts.append(ProfilerActivity.CUDA)
+            with profile(activities=acts) as prof:
+                loss_step = one_step()
+                if device.type == "cuda":
+                    torch.cuda.synchronize()
+            prof.export_chrome_trace(
+                os.path.join(config.rundir, f"trace_step0_rank{rank}.json"))
+        else:
+            loss_step = one_step()
 
         if it % 20 == 0:
             log_metrics(config, it, {"loss/optimized": loss_step})
