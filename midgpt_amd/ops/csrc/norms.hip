@@ -278,6 +278,9 @@ __global__ void qkv_prep_bwd_kernel(const u16* __restrict__ dq,
   float* smem = (float*)smem_raw;  // [2][C] partial dqw / dkw
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) smem[i] = 0.f;
   __syncthreads();
+  // per-thread dw accumulators (8 q-slots + 8 k-slots at this lane's sub
+  // offset) -> ONE atomic pass at the end instead of per-row atomics
+  float dwq_acc[8] = {0}, dwk_acc[8] = {0};
   for (long row = row0; row < nrows; row += rstep) {
     const int h = row % H;
     const int role = (row / H) % 3;
@@ -292,7 +295,7 @@ __global__ void qkv_prep_bwd_kernel(const u16* __restrict__ dq,
     const float* w = role == 0 ? qw : kw;
     const u16* dyp = role == 0 ? dq : dk;
     const float* stats = role == 0 ? qstats : kstats;
-    float* dwp = role == 0 ? smem : smem + C;
+    float* dw_acc = role == 0 ? dwq_acc : dwk_acc;
     const float mu = stats[2 * ((b * H + h) * T + t)];
     const float invstd = stats[2 * ((b * H + h) * T + t) + 1];
     u16x8 rawdy = *(const u16x8*)(dyp + in_off);
@@ -326,9 +329,14 @@ __global__ void qkv_prep_bwd_kernel(const u16* __restrict__ dq,
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       out[j] = f2b(invstd * (g[j] - mean_g - xh[j] * mean_gx));
-      atomicAdd(&dwp[sub * 8 + j], dn[j] * xh[j]);
+      dw_acc[j] += dn[j] * xh[j];
     }
     *(u16x8*)dst = out;
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&smem[sub * 8 + j], dwq_acc[j]);
+    atomicAdd(&smem[C + sub * 8 + j], dwk_acc[j]);
   }
   __syncthreads();
   for (int i = threadIdx.x; i < C; i += blockDim.x) {

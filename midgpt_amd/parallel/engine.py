@@ -101,11 +101,11 @@ class ShardedAdamW:
         in fp32 (reference microstep parity: src/train.py:85-92)."""
         if self.zero:
             pdist.reduce_scatter_flat(self.flat_g, self._g16_shard)
-            self.g32.add_(self._g16_shard.to(torch.float32))
+            self.g32.add_(self._g16_shard)
         else:
             if self.world > 1:
                 pdist.all_reduce_(self.flat_g)
-            self.g32.add_(self.flat_g.to(torch.float32))
+            self.g32.add_(self.flat_g)
         self.flat_g.zero_()
 
     # ------------------------------------------------------------------
