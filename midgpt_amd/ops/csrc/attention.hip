@@ -46,16 +46,23 @@ DEVINL void stage_rm(const u16* __restrict__ g, u16* lds) {
   }
 }
 // Transposed: global 32 x C (row stride C) -> LDS [C][32] swizzled.
+// Each thread loads TWO consecutive kv/q rows (u16x8 each) and writes
+// 8 ds_write_b32 pairs — consecutive kv are contiguous in the transposed
+// row, halving the LDS write instruction count vs scalar b16 stores.
 template <int C, int NT>
 DEVINL void stage_tr(const u16* __restrict__ g, u16* lds) {
 #pragma unroll
-  for (int idx = threadIdx.x * 8; idx < 32 * C; idx += NT * 8) {
-    const int row = idx / C, col = idx % C;  // row = kv/q index, col = c
-    u16x8 val = *(const u16x8*)(g + row * C + col);
+  for (int idx = threadIdx.x * 16; idx < 32 * C; idx += NT * 16) {
+    const int pair = idx / (2 * C);          // kv pair index (rows 2p, 2p+1)
+    const int col = (idx / 2) % C;           // c base (8 wide)
+    const int row = 2 * pair;
+    u16x8 v0 = *(const u16x8*)(g + (long)row * C + col);
+    u16x8 v1 = *(const u16x8*)(g + (long)(row + 1) * C + col);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int r = col + j;  // LDS row = c
-      *(u16*)((char*)lds + r * 64 + swz_tr(r, row * 2)) = val[j];
+      u16x2 pairv; pairv.x = v0[j]; pairv.y = v1[j];
+      *(u16x2*)((char*)lds + r * 64 + swz_tr(r, row * 2)) = pairv;
     }
   }
 }
@@ -238,8 +245,10 @@ __global__ void attn_delta_kernel(const u16* __restrict__ dO,
 }
 
 // ===========================================================================
-// Backward, kernel A (dK/dV): one WG = 128 k rows (32/wave); iterates q
-// tiles >= its diagonal. dK/dV accumulate in registers (no atomics).
+// Backward, kernel A (dK/dV): one WG = NW*32 k rows (32/wave); iterates q
+// tiles >= its diagonal with DOUBLE-BUFFERED staging (the tile qt+1 is
+// staged while qt is computed, hiding the HBM load latency that a
+// stage-barrier-compute structure exposes). dK/dV accumulate in registers.
 // S and dS are recomputed from Q,K,LSE (standard flash recompute).
 // ===========================================================================
 template <int C, int NW>
@@ -250,6 +259,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
     u16* __restrict__ dk, u16* __restrict__ dv, int B, int H, int T) {
   constexpr int NCB = C / 32;
   constexpr int NCH = C / 16;
+  constexpr int TILE = 4 * 32 * C;  // u16 elems per buffer set (Q,Qt,dO,dOt)
   const float scale = rsqrtf((float)C);
   const long bh = blockIdx.x % ((long)B * H);
   const int kb = blockIdx.x / (B * H);
@@ -259,12 +269,9 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
   const int myk = kw0 + (lane & 31);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  u16* ldsQ = (u16*)smem;              // [32*C] row-major
-  u16* ldsQt = ldsQ + 32 * C;          // [C*32] transposed
-  u16* ldsDO = ldsQt + C * 32;         // [32*C]
-  u16* ldsDOt = ldsDO + 32 * C;        // [C*32]
-  float* ldsLse = (float*)(ldsDOt + C * 32);
-  float* ldsDelta = ldsLse + 32;
+  u16* base = (u16*)smem;                       // [2][TILE]
+  float* ldsLse = (float*)(base + 2 * TILE);    // [2][32]
+  float* ldsDelta = ldsLse + 64;                // [2][32]
 
   const u16* qg = q + (bh * T) * C;
   const u16* kg = k + (bh * T) * C;
@@ -273,77 +280,96 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
   const u16* krow = kg + (long)myk * C;
   const u16* vrow = vg + (long)myk * C;
 
+  // wave-owned K row fragments; V re-read per tile (register budget)
+  bf16x8_t kf[NCH];
+#pragma unroll
+  for (int ch = 0; ch < NCH; ++ch)
+    kf[ch] = *(const bf16x8_t*)(krow + 16 * ch + 8 * (lane >> 5));
+
   f32x16 dvacc[NCB], dkacc[NCB];
 #pragma unroll
   for (int cb = 0; cb < NCB; ++cb) { dvacc[cb] = (f32x16)(0.f); dkacc[cb] = (f32x16)(0.f); }
 
   const int qt0 = kb * NW;  // diagonal q tile
   const int nqt = T / 32;
-  for (int qt = qt0; qt < nqt; ++qt) {
-    const int qbase = qt * 32;
-    __syncthreads();
-    stage_rm<C, NW * 64>(qg + (long)qbase * C, ldsQ);
-    stage_tr<C, NW * 64>(qg + (long)qbase * C, ldsQt);
-    stage_rm<C, NW * 64>(dog + (long)qbase * C, ldsDO);
-    stage_tr<C, NW * 64>(dog + (long)qbase * C, ldsDOt);
-    if (threadIdx.x < 32) {
-      ldsLse[threadIdx.x] = lse[bh * T + qbase + threadIdx.x];
-      ldsDelta[threadIdx.x] = delta[bh * T + qbase + threadIdx.x];
-    }
-    __syncthreads();
-    if (qbase + 31 < kw0) continue;  // fully masked for this wave
 
-    // S = Q x K^T (D rows = q regs, cols = k lanes); K frags from global
-    f32x16 s = (f32x16)(0.f);
-#pragma unroll
-    for (int ch = 0; ch < NCH; ++ch) {
-      bf16x8_t a = read_rm_frag<C>(ldsQ, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
-      bf16x8_t kf = *(const bf16x8_t*)(krow + 16 * ch + 8 * (lane >> 5));
-      s = mfma_32x32x16_bf16(a, kf, s);
+  auto stage_set = [&](int qt, int buf) {
+    const long qbase = (long)qt * 32;
+    u16* bq = base + buf * TILE;
+    stage_rm<C, NW * 64>(qg + qbase * C, bq);
+    stage_tr<C, NW * 64>(qg + qbase * C, bq + 32 * C);
+    stage_rm<C, NW * 64>(dog + qbase * C, bq + 2 * 32 * C);
+    stage_tr<C, NW * 64>(dog + qbase * C, bq + 3 * 32 * C);
+    if (threadIdx.x < 32) {
+      ldsLse[buf * 32 + threadIdx.x] = lse[bh * T + qbase + threadIdx.x];
+      ldsDelta[buf * 32 + threadIdx.x] = delta[bh * T + qbase + threadIdx.x];
     }
-    float p[16];
+  };
+
+  stage_set(qt0, 0);
+  __syncthreads();
+  for (int qt = qt0; qt < nqt; ++qt) {
+    const int buf = (qt - qt0) & 1;
+    if (qt + 1 < nqt) stage_set(qt + 1, buf ^ 1);
+    const int qbase = qt * 32;
+    if (qbase + 31 >= kw0) {  // not fully masked for this wave
+      const u16* ldsQ = base + buf * TILE;
+      const u16* ldsQt = ldsQ + 32 * C;
+      const u16* ldsDO = ldsQt + C * 32;
+      const u16* ldsDOt = ldsDO + 32 * C;
+      const float* lseb = ldsLse + buf * 32;
+      const float* deltab = ldsDelta + buf * 32;
+      // S = Q x K^T (D rows = q regs, cols = k lanes)
+      f32x16 s = (f32x16)(0.f);
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int qrow = qbase + mfma_d_row(lane, r);
-      p[r] = (myk > qrow) ? 0.f
-           : __expf(s[r] * scale - ldsLse[mfma_d_row(lane, r)]);
+      for (int ch = 0; ch < NCH; ++ch) {
+        bf16x8_t a = read_rm_frag<C>(ldsQ, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+        s = mfma_32x32x16_bf16(a, kf[ch], s);
+      }
+      float p[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = qbase + mfma_d_row(lane, r);
+        p[r] = (myk > qrow) ? 0.f
+             : __expf(s[r] * scale - lseb[mfma_d_row(lane, r)]);
+      }
+      // dV += P^T x dO
+      bf16x8_t pf0 = dlayout_to_afrag(p);
+      bf16x8_t pf1 = dlayout_to_afrag(p + 8);
+#pragma unroll
+      for (int cb = 0; cb < NCB; ++cb) {
+        bf16x8_t b0 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 16 * (lane >> 5));
+        bf16x8_t b1 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+        dvacc[cb] = mfma_32x32x16_bf16(pf0, b0, dvacc[cb]);
+        dvacc[cb] = mfma_32x32x16_bf16(pf1, b1, dvacc[cb]);
+      }
+      // dP = dO x V^T; V frags from global (L2-resident)
+      f32x16 dp = (f32x16)(0.f);
+#pragma unroll
+      for (int ch = 0; ch < NCH; ++ch) {
+        bf16x8_t a = read_rm_frag<C>(ldsDO, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+        bf16x8_t vf = *(const bf16x8_t*)(vrow + 16 * ch + 8 * (lane >> 5));
+        dp = mfma_32x32x16_bf16(a, vf, dp);
+      }
+      // dS = P * (dP - delta[q]) * scale; dK += dS^T x Q
+      float ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        ds[r] = p[r] * (dp[r] - deltab[mfma_d_row(lane, r)]) * scale;
+      bf16x8_t df0 = dlayout_to_afrag(ds);
+      bf16x8_t df1 = dlayout_to_afrag(ds + 8);
+#pragma unroll
+      for (int cb = 0; cb < NCB; ++cb) {
+        bf16x8_t b0 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 16 * (lane >> 5));
+        bf16x8_t b1 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+        dkacc[cb] = mfma_32x32x16_bf16(df0, b0, dkacc[cb]);
+        dkacc[cb] = mfma_32x32x16_bf16(df1, b1, dkacc[cb]);
+      }
     }
-    // dV += P^T x dO
-    bf16x8_t pf0 = dlayout_to_afrag(p);
-    bf16x8_t pf1 = dlayout_to_afrag(p + 8);
-#pragma unroll
-    for (int cb = 0; cb < NCB; ++cb) {
-      bf16x8_t b0 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 16 * (lane >> 5));
-      bf16x8_t b1 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
-      dvacc[cb] = mfma_32x32x16_bf16(pf0, b0, dvacc[cb]);
-      dvacc[cb] = mfma_32x32x16_bf16(pf1, b1, dvacc[cb]);
-    }
-    // dP = dO x V^T; V frags from global (L2-resident)
-    f32x16 dp = (f32x16)(0.f);
-#pragma unroll
-    for (int ch = 0; ch < NCH; ++ch) {
-      bf16x8_t a = read_rm_frag<C>(ldsDO, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
-      bf16x8_t vf = *(const bf16x8_t*)(vrow + 16 * ch + 8 * (lane >> 5));
-      dp = mfma_32x32x16_bf16(a, vf, dp);
-    }
-    // dS = P * (dP - delta[q]) * scale; dK += dS^T x Q
-    float ds[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r)
-      ds[r] = p[r] * (dp[r] - ldsDelta[mfma_d_row(lane, r)]) * scale;
-    bf16x8_t df0 = dlayout_to_afrag(ds);
-    bf16x8_t df1 = dlayout_to_afrag(ds + 8);
-#pragma unroll
-    for (int cb = 0; cb < NCB; ++cb) {
-      bf16x8_t b0 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 16 * (lane >> 5));
-      bf16x8_t b1 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
-      dkacc[cb] = mfma_32x32x16_bf16(df0, b0, dkacc[cb]);
-      dkacc[cb] = mfma_32x32x16_bf16(df1, b1, dkacc[cb]);
-    }
+    __syncthreads();
   }
 
   // epilogue: LDS bounce -> wide bf16 stores (reuses the staging region)
-  __syncthreads();
   float* ob = (float*)smem + w * 32 * 32;
   u16* dkg = dk + (bh * T + kw0) * C;
   u16* dvg = dv + (bh * T + kw0) * C;
@@ -381,9 +407,8 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
 }
 
 // ===========================================================================
-// Backward, kernel B (dQ): one WG = 128 q rows (32/wave); iterates k tiles
-// up to its diagonal (fwd-shaped traversal). dQ accumulates in registers,
-// direct bf16 store — no atomics.
+// Backward, kernel B (dQ): one WG = NW*32 q rows; iterates k tiles up to
+// its diagonal, DOUBLE-BUFFERED. dQ accumulates in registers — no atomics.
 // ===========================================================================
 template <int C, int NW>
 __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
@@ -393,6 +418,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
     u16* __restrict__ dq, int B, int H, int T) {
   constexpr int NCB = C / 32;
   constexpr int NCH = C / 16;
+  constexpr int TILE = 3 * 32 * C;  // u16 elems per buffer set (K, V, Kt)
   const float scale = rsqrtf((float)C);
   const long bh = blockIdx.x % ((long)B * H);
   const int qb = blockIdx.x / (B * H);
@@ -403,19 +429,15 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
   const int myq = qw0 + (lane & 31);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  u16* ldsK = (u16*)smem;          // [32*C] row-major
-  u16* ldsV = ldsK + 32 * C;       // [32*C] row-major
-  u16* ldsKt = ldsV + 32 * C;      // [C*32] transposed
-  u16* ldsDS = ldsKt + C * 32;     // per wave + w*32*32 (bf16)
+  u16* base = (u16*)smem;               // [2][TILE] = K rm | V rm | Kt
+  u16* ldsDS = base + 2 * TILE;         // per wave + w*32*32 (bf16)
 
   const u16* qg = q + (bh * T) * C;
   const u16* kg = k + (bh * T) * C;
   const u16* vg = v + (bh * T) * C;
   const u16* dog = dO + (bh * T) * C;
 
-  // per-lane Q A-fragments (row q = lane&31 within the wave tile); dO
-  // fragments are re-read per tile from global (L2-resident) to stay under
-  // the 256-VGPR spill cliff.
+  // per-lane Q A-fragments; dO fragments re-read per tile (register budget)
   bf16x8_t qf[NCH];
   const u16* dorow = dog + (long)myq * C;
   {
@@ -432,63 +454,72 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
   for (int cb = 0; cb < NCB; ++cb) dqacc[cb] = (f32x16)(0.f);
 
   const int nkt = (q0 + NW * 32) / 32;
+  auto stage_set = [&](int kt, int buf) {
+    const long k0 = (long)kt * 32;
+    u16* bk = base + buf * TILE;
+    stage_rm<C, NW * 64>(kg + k0 * C, bk);
+    stage_rm<C, NW * 64>(vg + k0 * C, bk + 32 * C);
+    stage_tr<C, NW * 64>(kg + k0 * C, bk + 2 * 32 * C);
+  };
+  stage_set(0, 0);
+  __syncthreads();
   for (int kt = 0; kt < nkt; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < nkt) stage_set(kt + 1, buf ^ 1);
     const int k0 = kt * 32;
-    __syncthreads();
-    stage_rm<C, NW * 64>(kg + (long)k0 * C, ldsK);
-    stage_rm<C, NW * 64>(vg + (long)k0 * C, ldsV);
-    stage_tr<C, NW * 64>(kg + (long)k0 * C, ldsKt);
-    __syncthreads();
-    if (k0 > qw0 + 31) continue;  // beyond this wave's diagonal
-
-    // S = Q x K^T and dP = dO x V^T in one pass.
-    // A = per-lane Q/dO fragments (A[q=lane&31][c]), B = K/V rows from LDS
-    // (B[c][k=lane&31]); D rows = q (reg-mapped via mfma_d_row), cols = k.
-    f32x16 s = (f32x16)(0.f);
-    f32x16 dp = (f32x16)(0.f);
+    if (k0 <= qw0 + 31) {
+      const u16* ldsK = base + buf * TILE;
+      const u16* ldsV = ldsK + 32 * C;
+      const u16* ldsKt = ldsV + 32 * C;
+      // S = Q x K^T and dP = dO x V^T in one pass.
+      // A = per-lane Q/dO fragments (A[q=lane&31][c]), B = K/V rows from LDS
+      // (B[c][k=lane&31]); D rows = q (reg-mapped), cols = k.
+      f32x16 s = (f32x16)(0.f);
+      f32x16 dp = (f32x16)(0.f);
 #pragma unroll
-    for (int ch = 0; ch < NCH; ++ch) {
-      bf16x8_t kfrag = read_rm_frag<C>(ldsK, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
-      s = mfma_32x32x16_bf16(qf[ch], kfrag, s);
-      bf16x8_t vfrag = read_rm_frag<C>(ldsV, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
-      bf16x8_t dof = *(const bf16x8_t*)(dorow + 16 * ch + 8 * (lane >> 5));
-      dp = mfma_32x32x16_bf16(dof, vfrag, dp);
-    }
-    // rows q are reg-mapped; cols k = lane&31. lse/delta per q row via shfl.
-    float ds[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int qrow = q0 + 32 * w + mfma_d_row(lane, r);
-      const float l_r = shfl32(mylse, mfma_d_row(lane, r));
-      const float d_r = shfl32(mydelta, mfma_d_row(lane, r));
-      const int kcol = k0 + (lane & 31);
-      float pv = (kcol > qrow) ? 0.f : __expf(s[r] * scale - l_r);
-      ds[r] = pv * (dp[r] - d_r) * scale;
-    }
-    // transpose dS through per-wave LDS -> A-frags A[q = lane&31][k]
-    u16* dsl = ldsDS + w * 32 * 32;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = mfma_d_row(lane, r);  // q-local
-      *(u16*)((char*)(dsl + row * 32) + swz_tr(row, (lane & 31) * 2)) = f2b(ds[r]);
-    }
-    __builtin_amdgcn_s_waitcnt(0);
-#pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      bf16x8_t a = *(const bf16x8_t*)((const char*)(dsl + (lane & 31) * 32) +
-                                      swz_tr(lane & 31, (16 * kc + 8 * (lane >> 5)) * 2));
-#pragma unroll
-      for (int cb = 0; cb < NCB; ++cb) {
-        bf16x8_t b = read_tr_frag(ldsKt, 32 * cb + (lane & 31),
-                                  (16 * kc + 8 * (lane >> 5)) * 2);
-        dqacc[cb] = mfma_32x32x16_bf16(a, b, dqacc[cb]);
+      for (int ch = 0; ch < NCH; ++ch) {
+        bf16x8_t kfrag = read_rm_frag<C>(ldsK, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+        s = mfma_32x32x16_bf16(qf[ch], kfrag, s);
+        bf16x8_t vfrag = read_rm_frag<C>(ldsV, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+        bf16x8_t dof = *(const bf16x8_t*)(dorow + 16 * ch + 8 * (lane >> 5));
+        dp = mfma_32x32x16_bf16(dof, vfrag, dp);
       }
+      // rows q are reg-mapped; cols k = lane&31. lse/delta per q via shfl.
+      float ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = qw0 + mfma_d_row(lane, r);
+        const float l_r = shfl32(mylse, mfma_d_row(lane, r));
+        const float d_r = shfl32(mydelta, mfma_d_row(lane, r));
+        const int kcol = k0 + (lane & 31);
+        float pv = (kcol > qrow) ? 0.f : __expf(s[r] * scale - l_r);
+        ds[r] = pv * (dp[r] - d_r) * scale;
+      }
+      // transpose dS through per-wave LDS -> A-frags A[q = lane&31][k]
+      u16* dsl = ldsDS + w * 32 * 32;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = mfma_d_row(lane, r);  // q-local
+        *(u16*)((char*)(dsl + row * 32) + swz_tr(row, (lane & 31) * 2)) = f2b(ds[r]);
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        bf16x8_t a = *(const bf16x8_t*)((const char*)(dsl + (lane & 31) * 32) +
+                                        swz_tr(lane & 31, (16 * kc + 8 * (lane >> 5)) * 2));
+#pragma unroll
+        for (int cb = 0; cb < NCB; ++cb) {
+          bf16x8_t b = read_tr_frag(ldsKt, 32 * cb + (lane & 31),
+                                    (16 * kc + 8 * (lane >> 5)) * 2);
+          dqacc[cb] = mfma_32x32x16_bf16(a, b, dqacc[cb]);
+        }
+      }
+      __builtin_amdgcn_s_waitcnt(0);  // dsl reads done before next overwrite
     }
-    __builtin_amdgcn_s_waitcnt(0);  // dsl reads done before next overwrite
+    __syncthreads();
   }
 
   // epilogue: LDS bounce -> wide stores
-  __syncthreads();
   float* ob = (float*)smem + w * 32 * 32;
   u16* dqg = dq + (bh * T + qw0) * C;
 #pragma unroll

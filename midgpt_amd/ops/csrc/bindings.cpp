@@ -227,12 +227,23 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
   auto dv = torch::empty_like(v);
   const int NW = (T % 256 == 0) ? 8 : 4;
   long grid = (long)B * H * (T / (NW * 32));
-  size_t smem_a = std::max((size_t)(4 * 32 * C) * 2 + 64 * 4,
+  // double-buffered staging: dkv = 2x(Q,Qt,dO,dOt) + lse/delta;
+  // dq = 2x(K,V,Kt) + per-wave dS. Above the 64 KiB default dynamic-LDS
+  // cap (gfx950 has 160 KiB/CU) -> raise the attribute.
+  size_t smem_a = std::max((size_t)(2 * 4 * 32 * C) * 2 + 2 * 64 * 4,
                            (size_t)(NW * 32 * 32 * 4));
-  size_t smem_b = std::max((size_t)(2 * 32 * C + C * 32) * 2 + NW * 32 * 32 * 2,
+  size_t smem_b = std::max((size_t)(2 * 3 * 32 * C) * 2 + NW * 32 * 32 * 2,
                            (size_t)(NW * 32 * 32 * 4));
 #define LAUNCH_BWD(CC, NN)                                                      \
   do {                                                                          \
+    if (smem_a > 64 * 1024)                                                     \
+      hipFuncSetAttribute(                                                      \
+          reinterpret_cast<const void*>(&attn_bwd_dkv_kernel<CC, NN>),          \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem_a);             \
+    if (smem_b > 64 * 1024)                                                     \
+      hipFuncSetAttribute(                                                      \
+          reinterpret_cast<const void*>(&attn_bwd_dq_kernel<CC, NN>),           \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem_b);             \
     hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NN>), dim3(grid), dim3(NN * 64),\
                        smem_a, cur_stream(), (const u16*)dO.data_ptr(),         \
                        (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),      \
