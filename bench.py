@@ -138,7 +138,7 @@ def main():
                 "g_accum_iters": g_accum,
                 "remat": config.remat,
                 "mfu_vs_2.5pf_dense": round(mfu, 4),
-                "last_loss": round(float(loss), 4),
+                "last_loss": round(float(loss.detach()), 4),
             },
         }), flush=True)
 
