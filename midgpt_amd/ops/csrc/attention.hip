@@ -33,38 +33,87 @@ DEVINL int swz_tr(int row, int byte_in_row) {
 }
 
 // ---------------------------------------------------------------------------
-// Cooperative staging helpers (256 threads)
+// Cooperative staging (T14 async split: issue global loads EARLY, hold the
+// tile in registers through the compute phase, ds_write late — the HBM
+// latency hides under the MFMA phase instead of stalling the wave at a
+// vmcnt before its LDS writes).
 // ---------------------------------------------------------------------------
-// Row-major 32 x C tile from global (row stride C) into swizzled LDS.
+// Single-row loader -> row-major swizzled image only.
 template <int C, int NT>
-DEVINL void stage_rm(const u16* __restrict__ g, u16* lds) {
+struct RmStage {
+  static constexpr int NV = (32 * C + NT * 8 - 1) / (NT * 8);
+  u16x8 v[NV];
+  DEVINL void load(const u16* __restrict__ g) {
 #pragma unroll
-  for (int idx = threadIdx.x * 8; idx < 32 * C; idx += NT * 8) {
-    const int row = idx / C, col = idx % C;
-    u16x8 val = *(const u16x8*)(g + row * C + col);
-    *(u16x8*)((char*)lds + row * C * 2 + swz_rm<C>(row, col * 2)) = val;
-  }
-}
-// Transposed: global 32 x C (row stride C) -> LDS [C][32] swizzled.
-// Each thread loads TWO consecutive kv/q rows (u16x8 each) and writes
-// 8 ds_write_b32 pairs — consecutive kv are contiguous in the transposed
-// row, halving the LDS write instruction count vs scalar b16 stores.
-template <int C, int NT>
-DEVINL void stage_tr(const u16* __restrict__ g, u16* lds) {
-#pragma unroll
-  for (int idx = threadIdx.x * 16; idx < 32 * C; idx += NT * 16) {
-    const int pair = idx / (2 * C);          // kv pair index (rows 2p, 2p+1)
-    const int col = (idx / 2) % C;           // c base (8 wide)
-    const int row = 2 * pair;
-    u16x8 v0 = *(const u16x8*)(g + (long)row * C + col);
-    u16x8 v1 = *(const u16x8*)(g + (long)(row + 1) * C + col);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int r = col + j;  // LDS row = c
-      u16x2 pairv; pairv.x = v0[j]; pairv.y = v1[j];
-      *(u16x2*)((char*)lds + r * 64 + swz_tr(r, row * 2)) = pairv;
+    for (int i = 0; i < NV; ++i) {
+      const int idx = threadIdx.x * 8 + i * NT * 8;
+      if (NV == 1 || idx < 32 * C)
+        v[i] = *(const u16x8*)(g + (long)(idx / C) * C + idx % C);
     }
   }
+  DEVINL void write(u16* lds) const {
+#pragma unroll
+    for (int i = 0; i < NV; ++i) {
+      const int idx = threadIdx.x * 8 + i * NT * 8;
+      if (NV == 1 || idx < 32 * C) {
+        const int row = idx / C, col = idx % C;
+        *(u16x8*)((char*)lds + row * C * 2 + swz_rm<C>(row, col * 2)) = v[i];
+      }
+    }
+  }
+};
+// Row-pair loader -> transposed image (u16x2 pair writes), and optionally
+// also the row-major image (both from ONE set of global loads).
+template <int C, int NT>
+struct TrStage {
+  static constexpr int NP = (32 * C + NT * 16 - 1) / (NT * 16);
+  u16x8 a[NP], b[NP];  // rows 2p and 2p+1 at an 8-col span
+  DEVINL void load(const u16* __restrict__ g) {
+#pragma unroll
+    for (int i = 0; i < NP; ++i) {
+      const int idx = threadIdx.x * 16 + i * NT * 16;
+      if (NP == 1 && idx >= 32 * C) continue;
+      const int row = 2 * (idx / (2 * C));
+      const int col = (idx / 2) % C;
+      a[i] = *(const u16x8*)(g + (long)row * C + col);
+      b[i] = *(const u16x8*)(g + (long)(row + 1) * C + col);
+    }
+  }
+  DEVINL void write_tr(u16* lds) const {
+#pragma unroll
+    for (int i = 0; i < NP; ++i) {
+      const int idx = threadIdx.x * 16 + i * NT * 16;
+      if (NP == 1 && idx >= 32 * C) continue;
+      const int row = 2 * (idx / (2 * C));
+      const int col = (idx / 2) % C;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int r = col + j;  // LDS row = c
+        u16x2 pv; pv.x = a[i][j]; pv.y = b[i][j];
+        *(u16x2*)((char*)lds + r * 64 + swz_tr(r, row * 2)) = pv;
+      }
+    }
+  }
+  DEVINL void write_rm(u16* lds) const {
+#pragma unroll
+    for (int i = 0; i < NP; ++i) {
+      const int idx = threadIdx.x * 16 + i * NT * 16;
+      if (NP == 1 && idx >= 32 * C) continue;
+      const int row = 2 * (idx / (2 * C));
+      const int col = (idx / 2) % C;
+      *(u16x8*)((char*)lds + row * C * 2 + swz_rm<C>(row, col * 2)) = a[i];
+      *(u16x8*)((char*)lds + (row + 1) * C * 2 + swz_rm<C>(row + 1, col * 2)) = b[i];
+    }
+  }
+};
+// Synchronous convenience wrappers (prologue use).
+template <int C, int NT>
+DEVINL void stage_rm(const u16* __restrict__ g, u16* lds) {
+  RmStage<C, NT> st; st.load(g); st.write(lds);
+}
+template <int C, int NT>
+DEVINL void stage_tr(const u16* __restrict__ g, u16* lds) {
+  TrStage<C, NT> st; st.load(g); st.write_tr(lds);
 }
 
 // Read one A/B fragment (bf16x8) from a swizzled row-major [R][C] tile:
@@ -129,11 +178,14 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restr
   stage_tr<C, NW * 64>(vg, ldsVt);
   __syncthreads();
 
+  RmStage<C, NW * 64> kst;
+  TrStage<C, NW * 64> vst;
   for (int kt = 0; kt < nkt; ++kt) {
     const int buf = kt & 1;
-    if (kt + 1 < nkt) {
-      stage_rm<C, NW * 64>(kg + (long)(kt + 1) * 32 * C, ldsK + (1 - buf) * 32 * C);
-      stage_tr<C, NW * 64>(vg + (long)(kt + 1) * 32 * C, ldsVt + (1 - buf) * C * 32);
+    const bool pre = kt + 1 < nkt;
+    if (pre) {  // T14: issue next tile's loads before this tile's compute
+      kst.load(kg + (long)(kt + 1) * 32 * C);
+      vst.load(vg + (long)(kt + 1) * 32 * C);
     }
     const int k0 = kt * 32;
     if (k0 <= qw0 + 31) {  // wave-uniform: tile not fully masked for this wave
@@ -184,6 +236,10 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restr
         oacc[cb] = mfma_32x32x16_bf16(pf0, b0, oacc[cb]);
         oacc[cb] = mfma_32x32x16_bf16(pf1, b1, oacc[cb]);
       }
+    }
+    if (pre) {  // T14: LDS writes after compute (loads have landed by now)
+      kst.write(ldsK + (1 - buf) * 32 * C);
+      vst.write_tr(ldsVt + (1 - buf) * C * 32);
     }
     __syncthreads();
   }
@@ -308,9 +364,20 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
 
   stage_set(qt0, 0);
   __syncthreads();
+  TrStage<C, NW * 64> qst, dost;  // ONE load feeds both rm and tr images
+  float lse_r = 0.f, del_r = 0.f;
   for (int qt = qt0; qt < nqt; ++qt) {
     const int buf = (qt - qt0) & 1;
-    if (qt + 1 < nqt) stage_set(qt + 1, buf ^ 1);
+    const bool pre = qt + 1 < nqt;
+    if (pre) {  // T14 split: loads early
+      const long nb = (long)(qt + 1) * 32;
+      qst.load(qg + nb * C);
+      dost.load(dog + nb * C);
+      if (threadIdx.x < 32) {
+        lse_r = lse[bh * T + nb + threadIdx.x];
+        del_r = delta[bh * T + nb + threadIdx.x];
+      }
+    }
     const int qbase = qt * 32;
     if (qbase + 31 >= kw0) {  // not fully masked for this wave
       const u16* ldsQ = base + buf * TILE;
@@ -364,6 +431,17 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
         bf16x8_t b1 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
         dkacc[cb] = mfma_32x32x16_bf16(df0, b0, dkacc[cb]);
         dkacc[cb] = mfma_32x32x16_bf16(df1, b1, dkacc[cb]);
+      }
+    }
+    if (pre) {  // writes late
+      u16* bq = base + (buf ^ 1) * TILE;
+      qst.write_rm(bq);
+      qst.write_tr(bq + 32 * C);
+      dost.write_rm(bq + 2 * 32 * C);
+      dost.write_tr(bq + 3 * 32 * C);
+      if (threadIdx.x < 32) {
+        ldsLse[(buf ^ 1) * 32 + threadIdx.x] = lse_r;
+        ldsDelta[(buf ^ 1) * 32 + threadIdx.x] = del_r;
       }
     }
     __syncthreads();
@@ -463,9 +541,15 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
   };
   stage_set(0, 0);
   __syncthreads();
+  TrStage<C, NW * 64> kst;  // ONE load feeds K rm + Kt images
+  RmStage<C, NW * 64> vst;
   for (int kt = 0; kt < nkt; ++kt) {
     const int buf = kt & 1;
-    if (kt + 1 < nkt) stage_set(kt + 1, buf ^ 1);
+    const bool pre = kt + 1 < nkt;
+    if (pre) {
+      kst.load(kg + (long)(kt + 1) * 32 * C);
+      vst.load(vg + (long)(kt + 1) * 32 * C);
+    }
     const int k0 = kt * 32;
     if (k0 <= qw0 + 31) {
       const u16* ldsK = base + buf * TILE;
@@ -515,6 +599,12 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
         }
       }
       __builtin_amdgcn_s_waitcnt(0);  // dsl reads done before next overwrite
+    }
+    if (pre) {
+      u16* bk = base + (buf ^ 1) * TILE;
+      kst.write_rm(bk);
+      vst.write(bk + 32 * C);
+      kst.write_tr(bk + 2 * 32 * C);
     }
     __syncthreads();
   }
