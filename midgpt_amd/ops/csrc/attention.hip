@@ -129,51 +129,62 @@ DEVINL bf16x8_t read_tr_frag(const u16* lds, int row, int kbyte) {
 DEVINL float shfl32(float v, int src) { return __shfl(v, src, 32); }
 
 // ===========================================================================
-// Forward
+// Forward. SPW strips per wave: with SPW=2 each wave owns MIRRORED strips
+// (w and NW*SPW-1-w), so causal work per wave is exactly equal and the
+// staging barriers park nobody (the measured 50% SQ_WAIT_ANY at SPW=1 came
+// from low-diagonal waves waiting on high-diagonal ones).
 // ===========================================================================
-template <int C, int NW>
-__global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restrict__ q,
+template <int C, int NW, int SPW, int MINW>
+__global__ __launch_bounds__(NW * 64, MINW) void attn_fwd_kernel(const u16* __restrict__ q,
                                 const u16* __restrict__ k,
                                 const u16* __restrict__ v,
                                 u16* __restrict__ o, float* __restrict__ lse,
                                 int B, int H, int T) {
   constexpr int NCB = C / 32;   // 32-col c-blocks
   constexpr int NCH = C / 16;   // 16-deep mfma chunks
+  constexpr int NSTRIP = NW * SPW;
   const float scale = rsqrtf((float)C);
   // qb-outermost grid order: all q-blocks of one (b,h) land on the same
   // XCD (b%8 dispatch) for K/V L2 reuse (guide T1).
   const long bh = blockIdx.x % ((long)B * H);
   const int qb = blockIdx.x / (B * H);
-  const int q0 = qb * (NW * 32);
+  const int q0 = qb * (NSTRIP * 32);
   const int lane = lane_id();
   const int w = wave_id();
-  const int qw0 = q0 + 32 * w;          // this wave's first q row
-  const int myq = qw0 + (lane & 31);    // this lane's q row
+  int strip[SPW];
+  strip[0] = w;
+  if (SPW == 2) strip[1] = NSTRIP - 1 - w;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   u16* ldsK = (u16*)smem;                       // [2][32*C]
   u16* ldsVt = (u16*)(smem + 2 * 32 * C * 2);   // [2][C*32]
-  float* obuf = (float*)smem;                   // epilogue reuse: [4][32*32]
+  float* obuf = (float*)smem;                   // epilogue reuse: [NW][32*32]
 
   const u16* qg = q + (bh * T) * C;
   const u16* kg = k + (bh * T) * C;
   const u16* vg = v + (bh * T) * C;
 
-  // Q B-fragments straight from global to registers.
+  // Q B-fragments in registers for the LAST (busiest) strip only; earlier
+  // strips re-read their Q rows from global per tile (L2-resident, few
+  // tiles) to stay under the 256-VGPR cap at 2 waves/SIMD.
   bf16x8_t qf[NCH];
+  f32x16 oacc[SPW][NCB];
+  float m[SPW], lsum[SPW];
   {
-    const u16* qrow = qg + (long)myq * C;
+    const u16* qrow = qg + (long)(q0 + 32 * strip[SPW - 1] + (lane & 31)) * C;
 #pragma unroll
     for (int ch = 0; ch < NCH; ++ch)
       qf[ch] = *(const bf16x8_t*)(qrow + 16 * ch + 8 * (lane >> 5));
   }
-
-  f32x16 oacc[NCB];
 #pragma unroll
-  for (int cb = 0; cb < NCB; ++cb) oacc[cb] = (f32x16)(0.f);
-  float m = -1e30f, lsum = 0.f;
+  for (int sp = 0; sp < SPW; ++sp) {
+#pragma unroll
+    for (int cb = 0; cb < NCB; ++cb) oacc[sp][cb] = (f32x16)(0.f);
+    m[sp] = -1e30f;
+    lsum[sp] = 0.f;
+  }
 
-  const int nkt = (q0 + NW * 32) / 32;
+  const int nkt = (q0 + NSTRIP * 32) / 32;
   stage_rm<C, NW * 64>(kg, ldsK);
   stage_tr<C, NW * 64>(vg, ldsVt);
   __syncthreads();
@@ -188,14 +199,21 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restr
       vst.load(vg + (long)(kt + 1) * 32 * C);
     }
     const int k0 = kt * 32;
-    if (k0 <= qw0 + 31) {  // wave-uniform: tile not fully masked for this wave
+#pragma unroll
+    for (int sp = 0; sp < SPW; ++sp) {
+      const int qw0 = q0 + 32 * strip[sp];
+      const int myq = qw0 + (lane & 31);
+      if (k0 > qw0 + 31) continue;  // wave-uniform per strip
       // S = K x Q^T  (swapped: D rows = k, cols = q)
       f32x16 s = (f32x16)(0.f);
       const u16* kb = ldsK + buf * 32 * C;
+      const u16* qrow = qg + (long)myq * C;
 #pragma unroll
       for (int ch = 0; ch < NCH; ++ch) {
         bf16x8_t a = read_rm_frag<C>(kb, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
-        s = mfma_32x32x16_bf16(a, qf[ch], s);
+        bf16x8_t qfr = (sp == SPW - 1) ? qf[ch]
+            : *(const bf16x8_t*)(qrow + 16 * ch + 8 * (lane >> 5));
+        s = mfma_32x32x16_bf16(a, qfr, s);
       }
       // mask + tile row-max (per q = lane&31; halves merged via xor 32)
       float sv[16];
@@ -206,9 +224,9 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restr
         mt = fmaxf(mt, sv[r]);
       }
       mt = fmaxf(mt, __shfl_xor(mt, 32));
-      const float mn = fmaxf(m, mt);
-      const float alpha = __expf((m - mn) * scale);
-      m = mn;
+      const float mn = fmaxf(m[sp], mt);
+      const float alpha = __expf((m[sp] - mn) * scale);
+      m[sp] = mn;
       float p[16], psum = 0.f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -216,7 +234,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restr
         psum += p[r];
       }
       psum += __shfl_xor(psum, 32);
-      lsum = lsum * alpha + psum;
+      lsum[sp] = lsum[sp] * alpha + psum;
       // rescale O by alpha[q_of_reg]
       float arow[16];
 #pragma unroll
@@ -224,7 +242,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restr
 #pragma unroll
       for (int cb = 0; cb < NCB; ++cb)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) oacc[cb][r] *= arow[r];
+        for (int r = 0; r < 16; ++r) oacc[sp][cb][r] *= arow[r];
       // P -> bf16 A-fragments; PV
       bf16x8_t pf0 = dlayout_to_afrag(p);
       bf16x8_t pf1 = dlayout_to_afrag(p + 8);
@@ -233,8 +251,8 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restr
       for (int cb = 0; cb < NCB; ++cb) {
         bf16x8_t b0 = read_tr_frag(vb, 32 * cb + (lane & 31), 16 * (lane >> 5));
         bf16x8_t b1 = read_tr_frag(vb, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
-        oacc[cb] = mfma_32x32x16_bf16(pf0, b0, oacc[cb]);
-        oacc[cb] = mfma_32x32x16_bf16(pf1, b1, oacc[cb]);
+        oacc[sp][cb] = mfma_32x32x16_bf16(pf0, b0, oacc[sp][cb]);
+        oacc[sp][cb] = mfma_32x32x16_bf16(pf1, b1, oacc[sp][cb]);
       }
     }
     if (pre) {  // T14: LDS writes after compute (loads have landed by now)
@@ -244,37 +262,42 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_fwd_kernel(const u16* __restr
     __syncthreads();
   }
 
-  // epilogue: normalize, bounce through LDS, wide stores
-  const float rec = 1.f / lsum;
-  float rrow[16];
+  // epilogue: normalize, bounce through LDS, wide stores (per strip)
 #pragma unroll
-  for (int r = 0; r < 16; ++r) rrow[r] = shfl32(rec, mfma_d_row(lane, r));
-  if (lane < 32) lse[bh * T + myq] = m * scale + __logf(lsum);
-  float* ob = obuf + w * 32 * 32;
-  u16* og = o + (bh * T + qw0) * C;
+  for (int sp = 0; sp < SPW; ++sp) {
+    const int qw0 = q0 + 32 * strip[sp];
+    const int myq = qw0 + (lane & 31);
+    const float rec = 1.f / lsum[sp];
+    float rrow[16];
 #pragma unroll
-  for (int cb = 0; cb < NCB; ++cb) {
+    for (int r = 0; r < 16; ++r) rrow[r] = shfl32(rec, mfma_d_row(lane, r));
+    if (lane < 32) lse[bh * T + myq] = m[sp] * scale + __logf(lsum[sp]);
+    float* ob = obuf + w * 32 * 32;
+    u16* og = o + (bh * T + qw0) * C;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = mfma_d_row(lane, r);
-      *(float*)((char*)ob + row * 128 + (((lane & 31) * 4) ^ ((row & 7) << 4))) =
-          oacc[cb][r] * rrow[r];
+    for (int cb = 0; cb < NCB; ++cb) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = mfma_d_row(lane, r);
+        *(float*)((char*)ob + row * 128 + (((lane & 31) * 4) ^ ((row & 7) << 4))) =
+            oacc[sp][cb][r] * rrow[r];
+      }
+      __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt: LDS writes land (same wave)
+      const int row = lane & 31;
+      const int c16 = 16 * (lane >> 5);
+      float tmp[16];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        f32x4 t = *(const f32x4*)((char*)ob + row * 128 + (((c16 + 4 * i) * 4) ^ ((row & 7) << 4)));
+        tmp[4 * i] = t[0]; tmp[4 * i + 1] = t[1]; tmp[4 * i + 2] = t[2]; tmp[4 * i + 3] = t[3];
+      }
+      u16x8 out0, out1;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) { out0[j] = f2b(tmp[j]); out1[j] = f2b(tmp[8 + j]); }
+      *(u16x8*)(og + (long)row * C + 32 * cb + c16) = out0;
+      *(u16x8*)(og + (long)row * C + 32 * cb + c16 + 8) = out1;
+      __builtin_amdgcn_s_waitcnt(0);
     }
-    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt: LDS writes land (same wave)
-    const int row = lane & 31;
-    const int c16 = 16 * (lane >> 5);
-    float tmp[16];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      f32x4 t = *(const f32x4*)((char*)ob + row * 128 + (((c16 + 4 * i) * 4) ^ ((row & 7) << 4)));
-      tmp[4 * i] = t[0]; tmp[4 * i + 1] = t[1]; tmp[4 * i + 2] = t[2]; tmp[4 * i + 3] = t[3];
-    }
-    u16x8 out0, out1;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) { out0[j] = f2b(tmp[j]); out1[j] = f2b(tmp[8 + j]); }
-    *(u16x8*)(og + (long)row * C + 32 * cb + c16) = out0;
-    *(u16x8*)(og + (long)row * C + 32 * cb + c16 + 8) = out1;
-    __builtin_amdgcn_s_waitcnt(0);
   }
 }
 
@@ -485,26 +508,29 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
 }
 
 // ===========================================================================
-// Backward, kernel B (dQ): one WG = NW*32 q rows; iterates k tiles up to
-// its diagonal, DOUBLE-BUFFERED. dQ accumulates in registers — no atomics.
+// Backward, kernel B (dQ): SPW mirrored q strips per wave (work balance,
+// see forward); iterates k tiles up to the max diagonal, DOUBLE-BUFFERED.
+// dQ accumulates in registers — no atomics.
 // ===========================================================================
-template <int C, int NW>
-__global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
+template <int C, int NW, int SPW, int MINW>
+__global__ __launch_bounds__(NW * 64, MINW) void attn_bwd_dq_kernel(
     const u16* __restrict__ dO, const u16* __restrict__ q,
     const u16* __restrict__ k, const u16* __restrict__ v,
     const float* __restrict__ lse, const float* __restrict__ delta,
     u16* __restrict__ dq, int B, int H, int T) {
   constexpr int NCB = C / 32;
   constexpr int NCH = C / 16;
+  constexpr int NSTRIP = NW * SPW;
   constexpr int TILE = 3 * 32 * C;  // u16 elems per buffer set (K, V, Kt)
   const float scale = rsqrtf((float)C);
   const long bh = blockIdx.x % ((long)B * H);
   const int qb = blockIdx.x / (B * H);
-  const int q0 = qb * (NW * 32);
+  const int q0 = qb * (NSTRIP * 32);
   const int lane = lane_id();
   const int w = wave_id();
-  const int qw0 = q0 + 32 * w;
-  const int myq = qw0 + (lane & 31);
+  int strip[SPW];
+  strip[0] = w;
+  if (SPW == 2) strip[1] = NSTRIP - 1 - w;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   u16* base = (u16*)smem;               // [2][TILE] = K rm | V rm | Kt
@@ -515,23 +541,25 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
   const u16* vg = v + (bh * T) * C;
   const u16* dog = dO + (bh * T) * C;
 
-  // per-lane Q A-fragments; dO fragments re-read per tile (register budget)
-  bf16x8_t qf[NCH];
-  const u16* dorow = dog + (long)myq * C;
-  {
-    const u16* qrow = qg + (long)myq * C;
+  bf16x8_t qf[NCH];  // registers only for the last (busiest) strip
+  f32x16 dqacc[SPW][NCB];
+  float mylse[SPW], mydelta[SPW];
 #pragma unroll
-    for (int ch = 0; ch < NCH; ++ch)
-      qf[ch] = *(const bf16x8_t*)(qrow + 16 * ch + 8 * (lane >> 5));
+  for (int sp = 0; sp < SPW; ++sp) {
+    const long myq = q0 + 32 * strip[sp] + (lane & 31);
+    if (sp == SPW - 1) {
+      const u16* qrow = qg + myq * C;
+#pragma unroll
+      for (int ch = 0; ch < NCH; ++ch)
+        qf[ch] = *(const bf16x8_t*)(qrow + 16 * ch + 8 * (lane >> 5));
+    }
+    mylse[sp] = lse[bh * T + myq];
+    mydelta[sp] = delta[bh * T + myq];
+#pragma unroll
+    for (int cb = 0; cb < NCB; ++cb) dqacc[sp][cb] = (f32x16)(0.f);
   }
-  const float mylse = lse[bh * T + myq];
-  const float mydelta = delta[bh * T + myq];
 
-  f32x16 dqacc[NCB];
-#pragma unroll
-  for (int cb = 0; cb < NCB; ++cb) dqacc[cb] = (f32x16)(0.f);
-
-  const int nkt = (q0 + NW * 32) / 32;
+  const int nkt = (q0 + NSTRIP * 32) / 32;
   auto stage_set = [&](int kt, int buf) {
     const long k0 = (long)kt * 32;
     u16* bk = base + buf * TILE;
@@ -551,19 +579,24 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
       vst.load(vg + (long)(kt + 1) * 32 * C);
     }
     const int k0 = kt * 32;
-    if (k0 <= qw0 + 31) {
-      const u16* ldsK = base + buf * TILE;
-      const u16* ldsV = ldsK + 32 * C;
-      const u16* ldsKt = ldsV + 32 * C;
+    const u16* ldsK = base + buf * TILE;
+    const u16* ldsV = ldsK + 32 * C;
+    const u16* ldsKt = ldsV + 32 * C;
+#pragma unroll
+    for (int sp = 0; sp < SPW; ++sp) {
+      const int qw0 = q0 + 32 * strip[sp];
+      if (k0 > qw0 + 31) continue;
+      const u16* qrow = qg + (long)(qw0 + (lane & 31)) * C;
+      const u16* dorow = dog + (long)(qw0 + (lane & 31)) * C;
       // S = Q x K^T and dP = dO x V^T in one pass.
-      // A = per-lane Q/dO fragments (A[q=lane&31][c]), B = K/V rows from LDS
-      // (B[c][k=lane&31]); D rows = q (reg-mapped), cols = k.
       f32x16 s = (f32x16)(0.f);
       f32x16 dp = (f32x16)(0.f);
 #pragma unroll
       for (int ch = 0; ch < NCH; ++ch) {
         bf16x8_t kfrag = read_rm_frag<C>(ldsK, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
-        s = mfma_32x32x16_bf16(qf[ch], kfrag, s);
+        bf16x8_t qfr = (sp == SPW - 1) ? qf[ch]
+            : *(const bf16x8_t*)(qrow + 16 * ch + 8 * (lane >> 5));
+        s = mfma_32x32x16_bf16(qfr, kfrag, s);
         bf16x8_t vfrag = read_rm_frag<C>(ldsV, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
         bf16x8_t dof = *(const bf16x8_t*)(dorow + 16 * ch + 8 * (lane >> 5));
         dp = mfma_32x32x16_bf16(dof, vfrag, dp);
@@ -573,8 +606,8 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrow = qw0 + mfma_d_row(lane, r);
-        const float l_r = shfl32(mylse, mfma_d_row(lane, r));
-        const float d_r = shfl32(mydelta, mfma_d_row(lane, r));
+        const float l_r = shfl32(mylse[sp], mfma_d_row(lane, r));
+        const float d_r = shfl32(mydelta[sp], mfma_d_row(lane, r));
         const int kcol = k0 + (lane & 31);
         float pv = (kcol > qrow) ? 0.f : __expf(s[r] * scale - l_r);
         ds[r] = pv * (dp[r] - d_r) * scale;
@@ -595,7 +628,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
         for (int cb = 0; cb < NCB; ++cb) {
           bf16x8_t b = read_tr_frag(ldsKt, 32 * cb + (lane & 31),
                                     (16 * kc + 8 * (lane >> 5)) * 2);
-          dqacc[cb] = mfma_32x32x16_bf16(a, b, dqacc[cb]);
+          dqacc[sp][cb] = mfma_32x32x16_bf16(a, b, dqacc[sp][cb]);
         }
       }
       __builtin_amdgcn_s_waitcnt(0);  // dsl reads done before next overwrite
@@ -609,33 +642,37 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dq_kernel(
     __syncthreads();
   }
 
-  // epilogue: LDS bounce -> wide stores
-  float* ob = (float*)smem + w * 32 * 32;
-  u16* dqg = dq + (bh * T + qw0) * C;
+  // epilogue: LDS bounce -> wide stores (per strip)
 #pragma unroll
-  for (int cb = 0; cb < NCB; ++cb) {
+  for (int sp = 0; sp < SPW; ++sp) {
+    const int qw0 = q0 + 32 * strip[sp];
+    float* ob = (float*)smem + w * 32 * 32;
+    u16* dqg = dq + (bh * T + qw0) * C;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = mfma_d_row(lane, r);
-      *(float*)((char*)ob + row * 128 + (((lane & 31) * 4) ^ ((row & 7) << 4))) =
-          dqacc[cb][r];
+    for (int cb = 0; cb < NCB; ++cb) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = mfma_d_row(lane, r);
+        *(float*)((char*)ob + row * 128 + (((lane & 31) * 4) ^ ((row & 7) << 4))) =
+            dqacc[sp][cb][r];
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+      const int row = lane & 31;
+      const int c16 = 16 * (lane >> 5);
+      float tmp[16];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        f32x4 t = *(const f32x4*)((char*)ob + row * 128 +
+                                  (((c16 + 4 * i) * 4) ^ ((row & 7) << 4)));
+        tmp[4 * i] = t[0]; tmp[4 * i + 1] = t[1];
+        tmp[4 * i + 2] = t[2]; tmp[4 * i + 3] = t[3];
+      }
+      u16x8 o0, o1;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) { o0[j] = f2b(tmp[j]); o1[j] = f2b(tmp[8 + j]); }
+      *(u16x8*)(dqg + (long)row * C + 32 * cb + c16) = o0;
+      *(u16x8*)(dqg + (long)row * C + 32 * cb + c16 + 8) = o1;
+      __builtin_amdgcn_s_waitcnt(0);
     }
-    __builtin_amdgcn_s_waitcnt(0);
-    const int row = lane & 31;
-    const int c16 = 16 * (lane >> 5);
-    float tmp[16];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      f32x4 t = *(const f32x4*)((char*)ob + row * 128 +
-                                (((c16 + 4 * i) * 4) ^ ((row & 7) << 4)));
-      tmp[4 * i] = t[0]; tmp[4 * i + 1] = t[1];
-      tmp[4 * i + 2] = t[2]; tmp[4 * i + 3] = t[3];
-    }
-    u16x8 o0, o1;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) { o0[j] = f2b(tmp[j]); o1[j] = f2b(tmp[8 + j]); }
-    *(u16x8*)(dqg + (long)row * C + 32 * cb + c16) = o0;
-    *(u16x8*)(dqg + (long)row * C + 32 * cb + c16 + 8) = o1;
-    __builtin_amdgcn_s_waitcnt(0);
   }
 }

@@ -192,20 +192,26 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   TORCH_CHECK(C == 64 || C == 128, "attn: head dim 64 or 128 (got ", C, ")");
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, T}, q.options().dtype(torch::kFloat));
-  // 8-wave WGs (256 q rows) halve K/V re-read bandwidth vs 4-wave; fall
-  // back to 4 waves when T % 256 != 0.
-  const int NW = (T % 256 == 0) ? 8 : 4;
-  long grid = (long)B * H * (T / (NW * 32));
+  // Mirrored-strip scheduling (SPW=2): causal work per wave is equal, so
+  // staging barriers park nobody. C=128 runs 4-wave WGs at 1 wave/SIMD
+  // (whole 512-register file: 256 VGPR + ~190 AGPR accumulators);
+  // C=64 fits 8-wave WGs at 2 waves/SIMD. Fallbacks for small T.
+  int NW, SPW, MINW;
+  if (C == 128) { NW = 4; SPW = (T % 256 == 0) ? 2 : 1; MINW = SPW == 2 ? 1 : 2; }
+  else { NW = (T % 256 == 0) ? 8 : 4; SPW = (T % (NW * 64) == 0) ? 2 : 1; MINW = 2; }
+  long grid = (long)B * H * (T / (NW * 32 * SPW));
   size_t smem = std::max((size_t)(4 * 32 * C * 2), (size_t)(NW * 32 * 32 * 4));
-#define LAUNCH_FWD(CC, NN)                                                      \
-  hipLaunchKernelGGL((attn_fwd_kernel<CC, NN>), dim3(grid), dim3(NN * 64), smem,\
-                     cur_stream(), (const u16*)q.data_ptr(),                    \
+#define LAUNCH_FWD(CC, NN, SS, MM)                                              \
+  hipLaunchKernelGGL((attn_fwd_kernel<CC, NN, SS, MM>), dim3(grid),             \
+                     dim3(NN * 64),                                             \
+                     smem, cur_stream(), (const u16*)q.data_ptr(),              \
                      (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),        \
                      (u16*)o.data_ptr(), lse.data_ptr<float>(), B, H, T)
-  if (C == 128 && NW == 8) LAUNCH_FWD(128, 8);
-  else if (C == 128) LAUNCH_FWD(128, 4);
-  else if (NW == 8) LAUNCH_FWD(64, 8);
-  else LAUNCH_FWD(64, 4);
+  if (C == 128 && SPW == 2) LAUNCH_FWD(128, 4, 2, 1);
+  else if (C == 128) LAUNCH_FWD(128, 4, 1, 2);
+  else if (NW == 8 && SPW == 2) LAUNCH_FWD(64, 8, 2, 2);
+  else if (NW == 8) LAUNCH_FWD(64, 8, 1, 2);
+  else LAUNCH_FWD(64, 4, 1, 2);
 #undef LAUNCH_FWD
   launch_check();
   return {o, lse};
@@ -225,41 +231,49 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
-  const int NW = (T % 256 == 0) ? 8 : 4;
-  long grid = (long)B * H * (T / (NW * 32));
+  const int NW_A = (T % 256 == 0) ? 8 : 4;  // dkv geometry
+  int NW_B, SPW, MINW;                       // dq geometry (mirrored strips)
+  if (C == 128) { NW_B = 4; SPW = (T % 256 == 0) ? 2 : 1; MINW = SPW == 2 ? 1 : 2; }
+  else { NW_B = (T % 256 == 0) ? 8 : 4; SPW = (T % (NW_B * 64) == 0) ? 2 : 1; MINW = 2; }
+  long grid_a = (long)B * H * (T / (NW_A * 32));
+  long grid_b = (long)B * H * (T / (NW_B * 32 * SPW));
   // double-buffered staging: dkv = 2x(Q,Qt,dO,dOt) + lse/delta;
   // dq = 2x(K,V,Kt) + per-wave dS. Above the 64 KiB default dynamic-LDS
   // cap (gfx950 has 160 KiB/CU) -> raise the attribute.
   size_t smem_a = std::max((size_t)(2 * 4 * 32 * C) * 2 + 2 * 64 * 4,
-                           (size_t)(NW * 32 * 32 * 4));
-  size_t smem_b = std::max((size_t)(2 * 3 * 32 * C) * 2 + NW * 32 * 32 * 2,
-                           (size_t)(NW * 32 * 32 * 4));
-#define LAUNCH_BWD(CC, NN)                                                      \
+                           (size_t)(NW_A * 32 * 32 * 4));
+  size_t smem_b = std::max((size_t)(2 * 3 * 32 * C) * 2 + NW_B * 32 * 32 * 2,
+                           (size_t)(NW_B * 32 * 32 * 4));
+#define LAUNCH_BWD(CC, NA, NB, SS, MM)                                          \
   do {                                                                          \
     if (smem_a > 64 * 1024)                                                     \
       hipFuncSetAttribute(                                                      \
-          reinterpret_cast<const void*>(&attn_bwd_dkv_kernel<CC, NN>),          \
+          reinterpret_cast<const void*>(&attn_bwd_dkv_kernel<CC, NA>),          \
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem_a);             \
     if (smem_b > 64 * 1024)                                                     \
       hipFuncSetAttribute(                                                      \
-          reinterpret_cast<const void*>(&attn_bwd_dq_kernel<CC, NN>),           \
+          reinterpret_cast<const void*>(&attn_bwd_dq_kernel<CC, NB, SS, MM>),   \
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem_b);             \
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NN>), dim3(grid), dim3(NN * 64),\
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NA>), dim3(grid_a),             \
+                       dim3(NA * 64),                                           \
                        smem_a, cur_stream(), (const u16*)dO.data_ptr(),         \
                        (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),      \
                        (const u16*)v.data_ptr(), lse.data_ptr<float>(),         \
                        delta.data_ptr<float>(), (u16*)dk.data_ptr(),            \
                        (u16*)dv.data_ptr(), B, H, T);                           \
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<CC, NN>), dim3(grid), dim3(NN * 64), \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<CC, NB, SS, MM>), dim3(grid_b),      \
+                       dim3(NB * 64),                                           \
                        smem_b, cur_stream(), (const u16*)dO.data_ptr(),         \
                        (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),      \
                        (const u16*)v.data_ptr(), lse.data_ptr<float>(),         \
                        delta.data_ptr<float>(), (u16*)dq.data_ptr(), B, H, T);  \
   } while (0)
-  if (C == 128 && NW == 8) LAUNCH_BWD(128, 8);
-  else if (C == 128) LAUNCH_BWD(128, 4);
-  else if (NW == 8) LAUNCH_BWD(64, 8);
-  else LAUNCH_BWD(64, 4);
+  if (C == 128 && SPW == 2) LAUNCH_BWD(128, 8, 4, 2, 1);
+  else if (C == 128 && NW_A == 8) LAUNCH_BWD(128, 8, 4, 1, 2);
+  else if (C == 128) LAUNCH_BWD(128, 4, 4, 1, 2);
+  else if (NW_B == 8 && SPW == 2) LAUNCH_BWD(64, 8, 8, 2, 2);
+  else if (NW_B == 8) LAUNCH_BWD(64, 8, 8, 1, 2);
+  else LAUNCH_BWD(64, 4, 4, 1, 2);
 #undef LAUNCH_BWD
   launch_check();
   return {dq, dk, dv};
