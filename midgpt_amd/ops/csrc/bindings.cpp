@@ -196,9 +196,11 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   // staging barriers park nobody. C=128 runs 4-wave WGs at 1 wave/SIMD
   // (whole 512-register file: 256 VGPR + ~190 AGPR accumulators);
   // C=64 fits 8-wave WGs at 2 waves/SIMD. Fallbacks for small T.
-  int NW, SPW, MINW;
-  if (C == 128) { NW = 4; SPW = (T % 256 == 0) ? 2 : 1; MINW = SPW == 2 ? 1 : 2; }
-  else { NW = (T % 256 == 0) ? 8 : 4; SPW = (T % (NW * 64) == 0) ? 2 : 1; MINW = 2; }
+  // Measured best: 8-wave WGs at 2 waves/SIMD (WG co-residency beats both
+  // 4-wave low-bandwidth and the 1-wave/SIMD mirrored-strip variant —
+  // SQ_WAIT is memory-wait dominated, so TLP matters most).
+  const int NW = (T % 256 == 0) ? 8 : 4;
+  const int SPW = 1;
   long grid = (long)B * H * (T / (NW * 32 * SPW));
   size_t smem = std::max((size_t)(4 * 32 * C * 2), (size_t)(NW * 32 * 32 * 4));
 #define LAUNCH_FWD(CC, NN, SS, MM)                                              \
@@ -207,9 +209,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
                      smem, cur_stream(), (const u16*)q.data_ptr(),              \
                      (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),        \
                      (u16*)o.data_ptr(), lse.data_ptr<float>(), B, H, T)
-  if (C == 128 && SPW == 2) LAUNCH_FWD(128, 4, 2, 1);
+  if (C == 128 && NW == 8) LAUNCH_FWD(128, 8, 1, 2);
   else if (C == 128) LAUNCH_FWD(128, 4, 1, 2);
-  else if (NW == 8 && SPW == 2) LAUNCH_FWD(64, 8, 2, 2);
   else if (NW == 8) LAUNCH_FWD(64, 8, 1, 2);
   else LAUNCH_FWD(64, 4, 1, 2);
 #undef LAUNCH_FWD
@@ -232,9 +233,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
   const int NW_A = (T % 256 == 0) ? 8 : 4;  // dkv geometry
-  int NW_B, SPW, MINW;                       // dq geometry (mirrored strips)
-  if (C == 128) { NW_B = 4; SPW = (T % 256 == 0) ? 2 : 1; MINW = SPW == 2 ? 1 : 2; }
-  else { NW_B = (T % 256 == 0) ? 8 : 4; SPW = (T % (NW_B * 64) == 0) ? 2 : 1; MINW = 2; }
+  const int NW_B = NW_A;                     // dq geometry
+  const int SPW = 1;
   long grid_a = (long)B * H * (T / (NW_A * 32));
   long grid_b = (long)B * H * (T / (NW_B * 32 * SPW));
   // double-buffered staging: dkv = 2x(Q,Qt,dO,dOt) + lse/delta;
@@ -268,10 +268,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
                        (const u16*)v.data_ptr(), lse.data_ptr<float>(),         \
                        delta.data_ptr<float>(), (u16*)dq.data_ptr(), B, H, T);  \
   } while (0)
-  if (C == 128 && SPW == 2) LAUNCH_BWD(128, 8, 4, 2, 1);
-  else if (C == 128 && NW_A == 8) LAUNCH_BWD(128, 8, 4, 1, 2);
+  if (C == 128 && NW_A == 8) LAUNCH_BWD(128, 8, 8, 1, 2);
   else if (C == 128) LAUNCH_BWD(128, 4, 4, 1, 2);
-  else if (NW_B == 8 && SPW == 2) LAUNCH_BWD(64, 8, 8, 2, 2);
   else if (NW_B == 8) LAUNCH_BWD(64, 8, 8, 1, 2);
   else LAUNCH_BWD(64, 4, 4, 1, 2);
 #undef LAUNCH_BWD
