@@ -209,8 +209,11 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
                      smem, cur_stream(), (const u16*)q.data_ptr(),              \
                      (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),        \
                      (u16*)o.data_ptr(), lse.data_ptr<float>(), B, H, T)
-  if (C == 128 && NW == 8) LAUNCH_FWD(128, 8, 1, 2);
+  static const bool minw3 = getenv("MIDGPT_ATTN_FWD_MINW3") != nullptr;
+  if (C == 128 && NW == 8 && minw3) LAUNCH_FWD(128, 8, 1, 3);
+  else if (C == 128 && NW == 8) LAUNCH_FWD(128, 8, 1, 2);
   else if (C == 128) LAUNCH_FWD(128, 4, 1, 2);
+  else if (NW == 8 && minw3) LAUNCH_FWD(64, 8, 1, 3);
   else if (NW == 8) LAUNCH_FWD(64, 8, 1, 2);
   else LAUNCH_FWD(64, 4, 1, 2);
 #undef LAUNCH_FWD
@@ -232,7 +235,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
-  const int NW_A = (T % 256 == 0) ? 8 : 4;  // dkv geometry
+  static const bool dkv4 = getenv("MIDGPT_ATTN_DKV_NW4") != nullptr;
+  const int NW_A = dkv4 ? 4 : ((T % 256 == 0) ? 8 : 4);  // dkv geometry
   const int NW_B = NW_A;                     // dq geometry
   const int SPW = 1;
   long grid_a = (long)B * H * (T / (NW_A * 32));
