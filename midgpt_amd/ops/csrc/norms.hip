@@ -199,16 +199,18 @@ __global__ void qkv_prep_fwd_kernel(const u16* __restrict__ qkv,
   const int RPW = WAVE / LPR;            // rows per wave
   const int sub = lane_id() % LPR;       // lane's slot within its row
   const int rsub = lane_id() / LPR;      // which row of the wave
-  const long nrows = (long)B * T * H * 3;
-  const long row0 = ((long)blockIdx.x * (blockDim.x / WAVE) + wave_id()) * RPW + rsub;
-  const long rstep = (long)gridDim.x * (blockDim.x / WAVE) * RPW;
-  for (long row = row0; row < nrows; row += rstep) {
-    const int h = row % H;
-    const int role = (row / H) % 3;
-    const long t = (row / ((long)3 * H)) % T;
-    const long b = row / ((long)3 * H * T);
-    const u16* src = qkv + ((((b * T + t) * 3 + role) * H + h) * C) + sub * 8;
-    const long out_off = (((b * H + h) * T + t) * C) + sub * 8;
+  const unsigned nrows = (unsigned)(B * T * H * 3);
+  const unsigned row0 = ((unsigned)blockIdx.x * (blockDim.x / WAVE) + wave_id()) * RPW + rsub;
+  const unsigned rstep = (unsigned)gridDim.x * (blockDim.x / WAVE) * RPW;
+  for (unsigned row = row0; row < nrows; row += rstep) {
+    const unsigned h = row % (unsigned)H;
+    const unsigned rest = row / (unsigned)H;
+    const unsigned role = rest % 3u;
+    const unsigned bt = rest / 3u;           // = b*T + t
+    const unsigned t = bt % (unsigned)T;
+    const unsigned b = bt / (unsigned)T;
+    const u16* src = qkv + (((long)bt * 3 + role) * H + h) * C + sub * 8;
+    const long out_off = ((((long)b * H + h) * T + t) * C) + sub * 8;
     u16x8 raw = *(const u16x8*)src;
     if (role == 2) {  // V: straight transpose copy
       *(u16x8*)(v + out_off) = raw;
@@ -232,8 +234,8 @@ __global__ void qkv_prep_fwd_kernel(const u16* __restrict__ qkv,
     for (int o = LPR / 2; o > 0; o >>= 1) ss += __shfl_xor(ss, o, LPR);
     const float invstd = rsqrtf(ss / C + eps);
     if (sub == 0) {
-      stats[2 * ((b * H + h) * T + t)] = mu;
-      stats[2 * ((b * H + h) * T + t) + 1] = invstd;
+      stats[2 * (((long)b * H + h) * T + t)] = mu;
+      stats[2 * (((long)b * H + h) * T + t) + 1] = invstd;
     }
     // normalize + weight + RoPE (pairs p = 4*sub + 0..3)
     const float* srow = sin_t + t * (C / 2) + sub * 4;
@@ -271,9 +273,9 @@ __global__ void qkv_prep_bwd_kernel(const u16* __restrict__ dq,
   const int RPW = WAVE / LPR;
   const int sub = lane_id() % LPR;
   const int rsub = lane_id() / LPR;
-  const long nrows = (long)B * T * H * 3;
-  const long row0 = ((long)blockIdx.x * (blockDim.x / WAVE) + wave_id()) * RPW + rsub;
-  const long rstep = (long)gridDim.x * (blockDim.x / WAVE) * RPW;
+  const unsigned nrows = (unsigned)(B * T * H * 3);
+  const unsigned row0 = ((unsigned)blockIdx.x * (blockDim.x / WAVE) + wave_id()) * RPW + rsub;
+  const unsigned rstep = (unsigned)gridDim.x * (blockDim.x / WAVE) * RPW;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* smem = (float*)smem_raw;  // [2][C] partial dqw / dkw
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) smem[i] = 0.f;
@@ -281,13 +283,15 @@ __global__ void qkv_prep_bwd_kernel(const u16* __restrict__ dq,
   // per-thread dw accumulators (8 q-slots + 8 k-slots at this lane's sub
   // offset) -> ONE atomic pass at the end instead of per-row atomics
   float dwq_acc[8] = {0}, dwk_acc[8] = {0};
-  for (long row = row0; row < nrows; row += rstep) {
-    const int h = row % H;
-    const int role = (row / H) % 3;
-    const long t = (row / ((long)3 * H)) % T;
-    const long b = row / ((long)3 * H * T);
-    u16* dst = dqkv + ((((b * T + t) * 3 + role) * H + h) * C) + sub * 8;
-    const long in_off = (((b * H + h) * T + t) * C) + sub * 8;
+  for (unsigned row = row0; row < nrows; row += rstep) {
+    const unsigned h = row % (unsigned)H;
+    const unsigned rest = row / (unsigned)H;
+    const unsigned role = rest % 3u;
+    const unsigned bt = rest / 3u;           // = b*T + t
+    const unsigned t = bt % (unsigned)T;
+    const unsigned b = bt / (unsigned)T;
+    u16* dst = dqkv + (((long)bt * 3 + role) * H + h) * C + sub * 8;
+    const long in_off = ((((long)b * H + h) * T + t) * C) + sub * 8;
     if (role == 2) {
       *(u16x8*)dst = *(const u16x8*)(dv + in_off);
       continue;
@@ -296,10 +300,10 @@ __global__ void qkv_prep_bwd_kernel(const u16* __restrict__ dq,
     const u16* dyp = role == 0 ? dq : dk;
     const float* stats = role == 0 ? qstats : kstats;
     float* dw_acc = role == 0 ? dwq_acc : dwk_acc;
-    const float mu = stats[2 * ((b * H + h) * T + t)];
-    const float invstd = stats[2 * ((b * H + h) * T + t) + 1];
+    const float mu = stats[2 * (((long)b * H + h) * T + t)];
+    const float invstd = stats[2 * (((long)b * H + h) * T + t) + 1];
     u16x8 rawdy = *(const u16x8*)(dyp + in_off);
-    u16x8 rawx = *(const u16x8*)(qkv + ((((b * T + t) * 3 + role) * H + h) * C) + sub * 8);
+    u16x8 rawx = *(const u16x8*)(qkv + (((long)bt * 3 + role) * H + h) * C + sub * 8);
     const float* srow = sin_t + t * (C / 2) + sub * 4;
     const float* crow = cos_t + t * (C / 2) + sub * 4;
     f32x4 sn = *(const f32x4*)srow;
