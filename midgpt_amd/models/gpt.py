@@ -109,6 +109,10 @@ class MLP(nn.Module):
 
     def forward(self, x):
         with _scope("mlp"):
+            if (x.is_cuda and x.dtype == torch.bfloat16
+                    and ops.have_ext() and os.environ.get("MIDGPT_FORCE_REF") != "1"):
+                return self.dropout(ops.fused_mlp(x, self.c_fc.weight,
+                                                  self.c_proj.weight))
             return self.dropout(
                 self.c_proj(F.gelu(self.c_fc(x), approximate="tanh")))
 

@@ -281,6 +281,38 @@ def cross_entropy(logits, targets):
 # chain clip_by_global_norm(1.0) -> scale_by_adam -> add_decayed_weights(
 # wd/lr_peak) -> scale_by_schedule -> scale(-1)  (reference src/train.py:153-159).
 # ----------------------------------------------------------------------------
+# ----------------------------------------------------------------------------
+# hipBLASLt epilogue-fused MLP: gelu lives in the c_fc GEMM epilogue (fwd)
+# and the dgrad GEMM epilogue (bwd) — no separate elementwise gelu kernels
+# (plan K7). Numerics: hipBLASLt GELU is the tanh approximation, matching
+# the reference jax.nn.gelu default.
+# ----------------------------------------------------------------------------
+class _FusedMLP(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2d, w1, w2):
+        g, h = _C.linear_gelu_fwd(x2d, w1)
+        y = torch.mm(g, w2.t())
+        ctx.save_for_backward(x2d, w1, w2, h, g)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, w1, w2, h, g = ctx.saved_tensors
+        dy = dy.contiguous()
+        dh = _C.matmul_dgelu(dy, w2, h)      # dgrad GEMM + dgelu epilogue
+        dw2 = torch.mm(dy.t(), g)
+        dx = torch.mm(dh, w1)
+        dw1 = torch.mm(dh.t(), x2d)
+        return dx, dw1, dw2
+
+
+def fused_mlp(x, w1, w2):
+    """x (..., D) -> gelu(x @ w1.T) @ w2.T with GEMM-epilogue GELU."""
+    shp = x.shape
+    y = _FusedMLP.apply(x.reshape(-1, shp[-1]).contiguous(), w1, w2)
+    return y.reshape(*shp[:-1], w2.shape[0])
+
+
 def adamw_step(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
                v: torch.Tensor, out_bf16: torch.Tensor | None,
                *, lr: float, beta1: float, beta2: float, eps: float,

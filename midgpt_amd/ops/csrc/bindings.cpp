@@ -299,6 +299,8 @@ torch::Tensor probe_pack(torch::Tensor M, torch::Tensor B) {
   return D;
 }
 
+#include "blaslt.inc"
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("rmsnorm_fwd", &rmsnorm_fwd);
   mod.def("rmsnorm_bwd", &rmsnorm_bwd);
@@ -309,6 +311,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adamw_step", &adamw_step);
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_bwd", &attn_bwd);
+  mod.def("linear_gelu_fwd", &linear_gelu_fwd);
+  mod.def("matmul_dgelu", &matmul_dgelu);
   mod.def("probe_mfma", &probe_mfma);
   mod.def("probe_pack", &probe_pack);
 }

@@ -14,6 +14,7 @@ setup(
         CUDAExtension(
             name="midgpt_amd.ops._C",
             sources=["midgpt_amd/ops/csrc/ext.hip"],
+            libraries=["hipblaslt"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],

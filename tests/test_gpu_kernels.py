@@ -167,3 +167,28 @@ def test_adamw_step_gpu_matches_cpu():
     assert relerr(m.cpu(), cmm) < 1e-6
     assert relerr(v.cpu(), cv) < 1e-6
     assert torch.equal(out16.cpu(), co)
+
+
+def test_fused_mlp_matches_reference():
+    torch.manual_seed(8)
+    M, D = 512, 256
+    x = torch.randn(M, D, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    w1 = torch.randn(4 * D, D, device=DEV, dtype=torch.bfloat16,
+                     requires_grad=True) * 0.05
+    w2 = torch.randn(D, 4 * D, device=DEV, dtype=torch.bfloat16,
+                     requires_grad=True) * 0.05
+    w1.retain_grad(); w2.retain_grad()
+    y = ops.fused_mlp(x, w1, w2)
+    g = torch.randn_like(y)
+    (y.float() * g.float()).sum().backward()
+    # reference in fp32
+    x2 = x.detach().float().requires_grad_(True)
+    w12 = w1.detach().float().requires_grad_(True)
+    w22 = w2.detach().float().requires_grad_(True)
+    import torch.nn.functional as F
+    y2 = F.linear(F.gelu(F.linear(x2, w12), approximate="tanh"), w22)
+    (y2 * g.float()).sum().backward()
+    assert relerr(y, y2) < 2e-2, relerr(y, y2)
+    assert relerr(x.grad, x2.grad) < 3e-2
+    assert relerr(w1.grad, w12.grad) < 3e-2
+    assert relerr(w2.grad, w22.grad) < 3e-2
