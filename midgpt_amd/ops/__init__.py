@@ -282,15 +282,18 @@ def cross_entropy(logits, targets):
 # wd/lr_peak) -> scale_by_schedule -> scale(-1)  (reference src/train.py:153-159).
 # ----------------------------------------------------------------------------
 # ----------------------------------------------------------------------------
-# hipBLASLt epilogue-fused MLP: gelu lives in the c_fc GEMM epilogue (fwd)
-# and the dgrad GEMM epilogue (bwd) — no separate elementwise gelu kernels
-# (plan K7). Numerics: hipBLASLt GELU is the tanh approximation, matching
-# the reference jax.nn.gelu default.
+# hipBLASLt DGELU-fused MLP backward: the gelu backward lives in the dgrad
+# GEMM epilogue — no separate elementwise gelu-backward kernel (plan K7).
+# (GELU_AUX forward fusion has no gfx950 hipblaslt algorithms on ROCm 7.2 —
+# probed via _C.lt_probe — so the forward keeps torch's GEMM + gelu;
+# numerics: hipBLASLt GELU is the tanh approximation, matching the
+# reference jax.nn.gelu default.)
 # ----------------------------------------------------------------------------
 class _FusedMLP(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x2d, w1, w2):
-        g, h = _C.linear_gelu_fwd(x2d, w1)
+        h = torch.mm(x2d, w1.t())
+        g = torch.nn.functional.gelu(h, approximate="tanh")
         y = torch.mm(g, w2.t())
         ctx.save_for_backward(x2d, w1, w2, h, g)
         return y

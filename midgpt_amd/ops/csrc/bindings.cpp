@@ -312,6 +312,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_bwd", &attn_bwd);
   mod.def("linear_gelu_fwd", &linear_gelu_fwd);
+  mod.def("lt_probe", &lt_probe);
   mod.def("matmul_dgelu", &matmul_dgelu);
   mod.def("probe_mfma", &probe_mfma);
   mod.def("probe_pack", &probe_pack);
