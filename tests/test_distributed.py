@@ -57,7 +57,9 @@ def _worker(rank, world, init_file, zero, out_q, steps=3):
     full[engine.shard_off:engine.shard_off + engine.shard_size] = engine.master
     if engine.zero:
         dist.all_reduce(full)
-    out_q.put((rank, full[:engine.total].clone()))
+    # send as numpy bytes: torch tensors over mp queues use fd-passing,
+    # which races with worker exit (flaky rebuild_storage_fd failures)
+    out_q.put((rank, full[:engine.total].numpy().copy()))
     dist.destroy_process_group()
 
 
@@ -73,7 +75,7 @@ def test_two_rank_matches_single_process(zero, tmp_path):
     results = {}
     for _ in range(2):
         r, full = q.get()
-        results[r] = full
+        results[r] = torch.from_numpy(full)
     for p in procs:
         p.join(timeout=120)
         assert p.exitcode == 0
