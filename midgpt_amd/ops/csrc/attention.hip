@@ -330,7 +330,7 @@ __global__ void attn_delta_kernel(const u16* __restrict__ dO,
 // stage-barrier-compute structure exposes). dK/dV accumulate in registers.
 // S and dS are recomputed from Q,K,LSE (standard flash recompute).
 // ===========================================================================
-template <int C, int NW>
+template <int C, int NW, int ABLATE = 0>
 __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
     const u16* __restrict__ dO, const u16* __restrict__ q,
     const u16* __restrict__ k, const u16* __restrict__ v,
@@ -392,7 +392,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
   for (int qt = qt0; qt < nqt; ++qt) {
     const int buf = (qt - qt0) & 1;
     const bool pre = qt + 1 < nqt;
-    if (pre) {  // T14 split: loads early
+    if (ABLATE != 2 && pre) {  // T14 split: loads early
       const long nb = (long)(qt + 1) * 32;
       qst.load(qg + nb * C);
       dost.load(dog + nb * C);
@@ -402,7 +402,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
       }
     }
     const int qbase = qt * 32;
-    if (qbase + 31 >= kw0) {  // not fully masked for this wave
+    if (ABLATE != 1 && qbase + 31 >= kw0) {  // not fully masked for this wave
       const u16* ldsQ = base + buf * TILE;
       const u16* ldsQt = ldsQ + 32 * C;
       const u16* ldsDO = ldsQt + C * 32;
@@ -456,7 +456,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
         dkacc[cb] = mfma_32x32x16_bf16(df1, b1, dkacc[cb]);
       }
     }
-    if (pre) {  // writes late
+    if (ABLATE != 2 && pre) {  // writes late
       u16* bq = base + (buf ^ 1) * TILE;
       qst.write_rm(bq);
       qst.write_tr(bq + 32 * C);

@@ -252,19 +252,34 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
   do {                                                                          \
     if (smem_a > 64 * 1024)                                                     \
       hipFuncSetAttribute(                                                      \
-          reinterpret_cast<const void*>(&attn_bwd_dkv_kernel<CC, NA>),          \
+          reinterpret_cast<const void*>(&attn_bwd_dkv_kernel<CC, NA, 0>),       \
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem_a);             \
     if (smem_b > 64 * 1024)                                                     \
       hipFuncSetAttribute(                                                      \
           reinterpret_cast<const void*>(&attn_bwd_dq_kernel<CC, NB, SS, MM>),   \
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem_b);             \
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NA>), dim3(grid_a),             \
-                       dim3(NA * 64),                                           \
-                       smem_a, cur_stream(), (const u16*)dO.data_ptr(),         \
-                       (const u16*)q.data_ptr(), (const u16*)k.data_ptr(),      \
-                       (const u16*)v.data_ptr(), lse.data_ptr<float>(),         \
-                       delta.data_ptr<float>(), (u16*)dk.data_ptr(),            \
-                       (u16*)dv.data_ptr(), B, H, T);                           \
+    static const char* abl = getenv("MIDGPT_DKV_ABLATE");                      \
+    if (abl && abl[0] == '1')                                                   \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NA, 1>), dim3(grid_a),        \
+                         dim3(NA * 64), smem_a, cur_stream(),                   \
+                         (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),   \
+                         (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),    \
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),        \
+                         (u16*)dk.data_ptr(), (u16*)dv.data_ptr(), B, H, T);    \
+    else if (abl && abl[0] == '2')                                              \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NA, 2>), dim3(grid_a),        \
+                         dim3(NA * 64), smem_a, cur_stream(),                   \
+                         (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),   \
+                         (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),    \
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),        \
+                         (u16*)dk.data_ptr(), (u16*)dv.data_ptr(), B, H, T);    \
+    else                                                                        \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NA, 0>), dim3(grid_a),        \
+                         dim3(NA * 64), smem_a, cur_stream(),                   \
+                         (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),   \
+                         (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),    \
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),        \
+                         (u16*)dk.data_ptr(), (u16*)dv.data_ptr(), B, H, T);    \
     hipLaunchKernelGGL((attn_bwd_dq_kernel<CC, NB, SS, MM>), dim3(grid_b),      \
                        dim3(NB * 64),                                           \
                        smem_b, cur_stream(), (const u16*)dO.data_ptr(),         \
