@@ -93,6 +93,14 @@ def train(config: ExperimentConfig):
 
     eval_batches = 1 if config.debug else 200
     t0 = time.perf_counter()
+    pbar = None
+    if pdist.is_main():
+        try:
+            from tqdm import tqdm
+            pbar = tqdm(total=config.max_steps, initial=first_step,
+                        dynamic_ncols=True)
+        except ImportError:
+            pbar = None
     for it in range(first_step, config.max_steps):
         if it % config.eval_interval == 0:
             tl = evaluate(model, loader, local_bs, device, eval_batches)
@@ -139,12 +147,19 @@ def train(config: ExperimentConfig):
             log_metrics(config, it, {"loss/optimized": loss_step})
         if mngr is not None and mngr.should_save(it):
             mngr.save(it, engine)
-        if pdist.is_main() and it % 10 == 0:
+        if pdist.is_main():
             dt = time.perf_counter() - t0
             thpt = (it - first_step + 1) * config.batch_size * \
                 config.g_accum_iters / max(dt, 1e-9)
-            print(f"step {it}: loss {loss_step:.4f} lr {lr:.2e} "
-                  f"thpt {thpt:.1f} seq/s", flush=True)
+            if pbar is not None:
+                pbar.update(1)
+                pbar.set_postfix(loss=f"{loss_step:.4f}", lr=f"{lr:.2e}",
+                                 thpt=f"{thpt:.1f} seq/s")
+            elif it % 10 == 0:
+                print(f"step {it}: loss {loss_step:.4f} lr {lr:.2e} "
+                      f"thpt {thpt:.1f} seq/s", flush=True)
+    if pbar is not None:
+        pbar.close()
     if mngr is not None:
         mngr.save(config.max_steps - 1, engine)
         mngr.wait()
