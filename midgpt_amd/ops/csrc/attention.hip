@@ -47,7 +47,7 @@ struct RmStage {
 #pragma unroll
     for (int i = 0; i < NV; ++i) {
       const int idx = threadIdx.x * 8 + i * NT * 8;
-      if (NV == 1 || idx < 32 * C)
+      if (idx < 32 * C)
         v[i] = *(const u16x8*)(g + (long)(idx / C) * C + idx % C);
     }
   }
@@ -55,7 +55,7 @@ struct RmStage {
 #pragma unroll
     for (int i = 0; i < NV; ++i) {
       const int idx = threadIdx.x * 8 + i * NT * 8;
-      if (NV == 1 || idx < 32 * C) {
+      if (idx < 32 * C) {
         const int row = idx / C, col = idx % C;
         *(u16x8*)((char*)lds + row * C * 2 + swz_rm<C>(row, col * 2)) = v[i];
       }
@@ -72,7 +72,7 @@ struct TrStage {
 #pragma unroll
     for (int i = 0; i < NP; ++i) {
       const int idx = threadIdx.x * 16 + i * NT * 16;
-      if (NP == 1 && idx >= 32 * C) continue;
+      if (idx >= 32 * C) continue;
       const int row = 2 * (idx / (2 * C));
       const int col = (idx / 2) % C;
       a[i] = *(const u16x8*)(g + (long)row * C + col);
@@ -83,7 +83,7 @@ struct TrStage {
 #pragma unroll
     for (int i = 0; i < NP; ++i) {
       const int idx = threadIdx.x * 16 + i * NT * 16;
-      if (NP == 1 && idx >= 32 * C) continue;
+      if (idx >= 32 * C) continue;
       const int row = 2 * (idx / (2 * C));
       const int col = (idx / 2) % C;
 #pragma unroll
@@ -98,7 +98,7 @@ struct TrStage {
 #pragma unroll
     for (int i = 0; i < NP; ++i) {
       const int idx = threadIdx.x * 16 + i * NT * 16;
-      if (NP == 1 && idx >= 32 * C) continue;
+      if (idx >= 32 * C) continue;
       const int row = 2 * (idx / (2 * C));
       const int col = (idx / 2) % C;
       *(u16x8*)((char*)lds + row * C * 2 + swz_rm<C>(row, col * 2)) = a[i];
