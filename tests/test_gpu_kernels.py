@@ -95,7 +95,7 @@ def test_qkv_prep_fwd_bwd(B, T, H, C):
 
 
 @pytest.mark.parametrize("B,H,T,C", [(2, 2, 128, 64), (1, 2, 256, 128),
-                                     (2, 1, 1024, 128)])
+                                     (2, 1, 1024, 128), (2, 2, 1024, 64)])
 def test_attention_fwd(B, H, T, C):
     torch.manual_seed(4)
     q = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16)
@@ -112,7 +112,8 @@ def test_attention_fwd(B, H, T, C):
     assert relerr(o, o_ref) < 2e-2, relerr(o, o_ref)
 
 
-@pytest.mark.parametrize("B,H,T,C", [(2, 2, 128, 64), (1, 2, 256, 128)])
+@pytest.mark.parametrize("B,H,T,C", [(2, 2, 128, 64), (1, 2, 256, 128),
+                                     (1, 1, 1024, 64), (1, 1, 1024, 128)])
 def test_attention_bwd(B, H, T, C):
     torch.manual_seed(5)
     q = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16, requires_grad=True)
