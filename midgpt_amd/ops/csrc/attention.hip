@@ -226,6 +226,7 @@ __global__ __launch_bounds__(NW * 64, MINW) void attn_fwd_kernel(const u16* __re
       mt = fmaxf(mt, __shfl_xor(mt, 32));
       const float mn = fmaxf(m[sp], mt);
       const float alpha = __expf((m[sp] - mn) * scale);
+      const bool need_rescale = !__all(mt <= m[sp]);  // EXACT: alpha==1 otherwise
       m[sp] = mn;
       float p[16], psum = 0.f;
 #pragma unroll
@@ -235,14 +236,17 @@ __global__ __launch_bounds__(NW * 64, MINW) void attn_fwd_kernel(const u16* __re
       }
       psum += __shfl_xor(psum, 32);
       lsum[sp] = lsum[sp] * alpha + psum;
-      // rescale O by alpha[q_of_reg]
-      float arow[16];
+      // rescale O by alpha[q_of_reg] — skipped (exactly) when no row's max
+      // grew this tile, which is most tiles once the running max settles
+      if (need_rescale) {
+        float arow[16];
 #pragma unroll
-      for (int r = 0; r < 16; ++r) arow[r] = shfl32(alpha, mfma_d_row(lane, r));
+        for (int r = 0; r < 16; ++r) arow[r] = shfl32(alpha, mfma_d_row(lane, r));
 #pragma unroll
-      for (int cb = 0; cb < NCB; ++cb)
+        for (int cb = 0; cb < NCB; ++cb)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) oacc[sp][cb][r] *= arow[r];
+          for (int r = 0; r < 16; ++r) oacc[sp][cb][r] *= arow[r];
+      }
       // P -> bf16 A-fragments; PV
       bf16x8_t pf0 = dlayout_to_afrag(p);
       bf16x8_t pf1 = dlayout_to_afrag(p + 8);
