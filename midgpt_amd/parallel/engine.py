@@ -79,34 +79,100 @@ class ShardedAdamW:
         self.master = full32[self.shard_off:self.shard_off + self.shard_size].clone()
         self.flat_w = full32.to(compute_dtype)
         del full32
-        self.flat_g = torch.zeros(self.padded, dtype=compute_dtype, device=device)
-        # Rebind parameters and grads as views into the flat buffers.
+        # DOUBLE flat grad buffers on GPU: microstep i's reduction runs on a
+        # side HIP stream while microstep i+1's forward/backward fills the
+        # other buffer (comm/compute overlap). CPU keeps one buffer with
+        # synchronous collectives.
+        self._overlap = device.type == "cuda"
+        nbuf = 2 if self._overlap else 1
+        self.flat_gs = [torch.zeros(self.padded, dtype=compute_dtype,
+                                    device=device) for _ in range(nbuf)]
+        self._cur = 0
+        self.params = params
+        # Rebind parameters as views into the flat weight buffer; grads are
+        # bound to the current flat grad buffer by _rebind_grads.
         for p, (_, shape, o, n) in zip(params, self.meta):
             p.data = self.flat_w[o:o + n].view(shape)
-            p.grad = self.flat_g[o:o + n].view(shape)
-        self.params = params
+        self._rebind_grads()
 
         self.m = torch.zeros_like(self.master)
         self.v = torch.zeros_like(self.master)
         # fp32 microbatch-grad accumulator over the shard (zeros each step)
         self.g32 = torch.zeros_like(self.master)
-        self._g16_shard = torch.empty(self.shard_size, dtype=compute_dtype,
-                                      device=device) if self.zero else None
+        self._g16_shards = [torch.empty(self.shard_size, dtype=compute_dtype,
+                                        device=device) for _ in range(nbuf)] \
+            if self.zero else None
+        if self._overlap:
+            self._comm_stream = torch.cuda.Stream(device=device)
+            self._pending = []  # [(done_event, buf_index), ...]
+
+    @property
+    def flat_g(self):
+        """The flat grad buffer autograd currently accumulates into."""
+        return self.flat_gs[self._cur]
+
+    def _rebind_grads(self):
+        cur = self.flat_gs[self._cur]
+        for p, (_, shape, o, n) in zip(self.params, self.meta):
+            p.grad = cur[o:o + n].view(shape)
+
+    def _drain(self, buf: int | None = None):
+        """Wait for pending side-stream reductions (all, or just ``buf``'s)
+        and fold them into the fp32 accumulator on the compute stream."""
+        keep = []
+        for done, b in self._pending:
+            if buf is not None and b != buf:
+                keep.append((done, b))
+                continue
+            torch.cuda.current_stream().wait_event(done)
+            if self.zero:
+                self.g32.add_(self._g16_shards[b])
+            else:
+                self.g32.add_(self.flat_gs[b])
+            self.flat_gs[b].zero_()
+        self._pending = keep
 
     # ------------------------------------------------------------------
     # per-microstep: called after each microbatch backward
     # ------------------------------------------------------------------
     def microstep_end(self):
         """Reduce this microbatch's bf16 grads across ranks and accumulate
-        in fp32 (reference microstep parity: src/train.py:85-92)."""
-        if self.zero:
-            pdist.reduce_scatter_flat(self.flat_g, self._g16_shard)
-            self.g32.add_(self._g16_shard)
-        else:
-            if self.world > 1:
-                pdist.all_reduce_(self.flat_g)
-            self.g32.add_(self.flat_g)
-        self.flat_g.zero_()
+        in fp32 (reference microstep parity: src/train.py:85-92).
+
+        GPU: the collective is ENQUEUED on a side HIP stream and the grad
+        views swap to the other flat buffer, so the reduce-scatter overlaps
+        the next microbatch's forward/backward (the last microstep's
+        reduction is folded in at step()). The pipeline — events, buffer
+        swap, view rebinding, deferred accumulation — runs identically at
+        world=1, so single-GPU tests and the bench exercise the full path;
+        only the collective call itself needs more ranks."""
+        if not self._overlap:  # CPU: synchronous
+            if self.zero:
+                pdist.reduce_scatter_flat(self.flat_gs[0], self._g16_shards[0])
+                self.g32.add_(self._g16_shards[0])
+            else:
+                if self.world > 1:
+                    pdist.all_reduce_(self.flat_gs[0])
+                self.g32.add_(self.flat_gs[0])
+            self.flat_gs[0].zero_()
+            return
+        cur = self._cur
+        ready = torch.cuda.Event()
+        ready.record()  # grads of this microbatch complete (compute stream)
+        with torch.cuda.stream(self._comm_stream):
+            self._comm_stream.wait_event(ready)
+            if self.zero:
+                pdist.reduce_scatter_flat(self.flat_gs[cur],
+                                          self._g16_shards[cur])
+            elif self.world > 1:
+                pdist.all_reduce_(self.flat_gs[cur])
+            done = torch.cuda.Event()
+            done.record(self._comm_stream)
+        self._pending.append((done, cur))
+        # swap to the other buffer for the next microbatch
+        self._cur = cur ^ 1
+        self._drain(buf=self._cur)  # its previous reduction must be consumed
+        self._rebind_grads()
 
     # ------------------------------------------------------------------
     # per-step
@@ -115,6 +181,8 @@ class ShardedAdamW:
         """Apply one optimizer step. Returns a device scalar tensor from
         which the pre-clip global grad norm is ``sqrt(t) * scale`` (no host
         sync on the step path). ``lr`` is this step's scheduled LR."""
+        if self._overlap:
+            self._drain()
         scale = 1.0 / (g_accum_iters * self.world)
         sq = self.g32.pow(2).sum()
         if self.zero:
