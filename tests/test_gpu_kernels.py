@@ -193,3 +193,24 @@ def test_fused_mlp_matches_reference():
     assert relerr(x.grad, x2.grad) < 3e-2
     assert relerr(w1.grad, w12.grad) < 3e-2
     assert relerr(w2.grad, w22.grad) < 3e-2
+
+
+def test_attention_fwd_spiked_max():
+    """Force late running-max growth (guide rule 26): a spiked K row in the
+    LAST kv tile makes every q row's max jump there, exercising the
+    rescale branch of the online softmax on the final tile."""
+    torch.manual_seed(9)
+    B, H, T, C = 1, 2, 512, 128
+    q = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16)
+    # spike: K row near the end aligned with every q (q . k_spike >> others)
+    k[:, :, T - 3] = (q.mean(dim=2) * 8).clamp(-64, 64)
+    o, lse = ops._C.attn_fwd(q, k, v)
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) / math.sqrt(C)
+    mask = torch.ones(T, T, dtype=torch.bool, device=DEV).tril()
+    s = s.masked_fill(~mask, float("-inf"))
+    lse_ref = torch.logsumexp(s, dim=-1)
+    o_ref = torch.matmul(torch.softmax(s, -1), v.float())
+    assert relerr(lse, lse_ref) < 1e-3
+    assert relerr(o, o_ref) < 3e-2, relerr(o, o_ref)
