@@ -128,55 +128,6 @@ __global__ void rmsnorm_bwd_bf16(const u16* __restrict__ dy, const u16* __restri
   }
 }
 
-// dx = r*g - x * r^3 * mean(x*g), g = dy*w
-template <typename T, int VEC>
-__global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
-                                   const float* __restrict__ w,
-                                   const float* __restrict__ invrms,
-                                   T* __restrict__ dx, float* __restrict__ dw_partial,
-                                   long N, int D) {
-  const int lane = lane_id();
-  const long row0 = (long)blockIdx.x * (blockDim.x / WAVE) + wave_id();
-  const long rstep = (long)gridDim.x * (blockDim.x / WAVE);
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* smem = (float*)smem_raw;  // dw partials: one (D,) slab per block
-  if (dw_partial) {
-    for (int i = threadIdx.x; i < D; i += blockDim.x) smem[i] = 0.f;
-    __syncthreads();
-  }
-  for (long row = row0; row < N; row += rstep) {
-    const T* dyr = dy + row * D;
-    const T* xr = x + row * D;
-    const float r = invrms[row];
-    float dot = 0.f;
-    for (int i = lane * VEC; i < D; i += WAVE * VEC) {
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) {
-        float g = load_as_f32<T>(dyr, i + j);
-        if (w) g *= w[i + j];
-        dot += load_as_f32<T>(xr, i + j) * g;
-      }
-    }
-    dot = group_sum<WAVE>(dot) / D;
-    T* dxr = dx + row * D;
-    for (int i = lane * VEC; i < D; i += WAVE * VEC) {
-#pragma unroll
-      for (int j = 0; j < VEC; ++j) {
-        float g = load_as_f32<T>(dyr, i + j);
-        float xv = load_as_f32<T>(xr, i + j);
-        float gw = w ? g * w[i + j] : g;
-        store_from_f32<T>(dxr, i + j, r * gw - xv * r * r * r * dot);
-        if (dw_partial) atomicAdd(&smem[i + j], g * xv * r);
-      }
-    }
-  }
-  if (dw_partial) {
-    __syncthreads();
-    for (int i = threadIdx.x; i < D; i += blockDim.x)
-      dw_partial[(long)blockIdx.x * D + i] = smem[i];
-  }
-}
-
 // ============================================================
 // Fused QK-LayerNorm + RoPE over packed QKV (K3+K4).
 // qkv: (B, T, 3, H, C) bf16 -> q,k (B,H,T,C) LN'd + RoPE'd, v transposed.

@@ -26,9 +26,6 @@ clip_by_global_norm(1.0) -> Adam(b1=.9, b2=cfg) -> + (wd/lr_peak)*theta
 """
 from __future__ import annotations
 
-import math
-from typing import Iterator
-
 import torch
 
 from midgpt_amd import ops
