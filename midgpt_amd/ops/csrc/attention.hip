@@ -424,16 +424,29 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrow = qbase + mfma_d_row(lane, r);
-        p[r] = (myk > qrow) ? 0.f
-             : __expf(s[r] * scale - lseb[mfma_d_row(lane, r)]);
+        if (ABLATE == 3)
+          p[r] = s[r] * scale - lseb[mfma_d_row(lane, r)];  // no exp/mask
+        else
+          p[r] = (myk > qrow) ? 0.f
+               : __expf(s[r] * scale - lseb[mfma_d_row(lane, r)]);
       }
       // dV += P^T x dO
-      bf16x8_t pf0 = dlayout_to_afrag(p);
-      bf16x8_t pf1 = dlayout_to_afrag(p + 8);
+      bf16x8_t pf0, pf1;
+      if (ABLATE == 4) {  // skip the permlane pack, keep p live
+        asm volatile("" :: "v"(p[0]), "v"(p[8]));
+        pf0 = kf[0]; pf1 = kf[1];
+      } else {
+        pf0 = dlayout_to_afrag(p);
+        pf1 = dlayout_to_afrag(p + 8);
+      }
 #pragma unroll
       for (int cb = 0; cb < NCB; ++cb) {
-        bf16x8_t b0 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 16 * (lane >> 5));
-        bf16x8_t b1 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+        bf16x8_t b0, b1;
+        if (ABLATE == 5) { b0 = kf[0]; b1 = kf[1]; }  // skip LDS frag reads
+        else {
+          b0 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 16 * (lane >> 5));
+          b1 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+        }
         dvacc[cb] = mfma_32x32x16_bf16(pf0, b0, dvacc[cb]);
         dvacc[cb] = mfma_32x32x16_bf16(pf1, b1, dvacc[cb]);
       }
@@ -450,12 +463,22 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r)
         ds[r] = p[r] * (dp[r] - deltab[mfma_d_row(lane, r)]) * scale;
-      bf16x8_t df0 = dlayout_to_afrag(ds);
-      bf16x8_t df1 = dlayout_to_afrag(ds + 8);
+      bf16x8_t df0, df1;
+      if (ABLATE == 4) {
+        asm volatile("" :: "v"(ds[0]), "v"(ds[8]));
+        df0 = kf[0]; df1 = kf[1];
+      } else {
+        df0 = dlayout_to_afrag(ds);
+        df1 = dlayout_to_afrag(ds + 8);
+      }
 #pragma unroll
       for (int cb = 0; cb < NCB; ++cb) {
-        bf16x8_t b0 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 16 * (lane >> 5));
-        bf16x8_t b1 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+        bf16x8_t b0, b1;
+        if (ABLATE == 5) { b0 = kf[0]; b1 = kf[1]; }
+        else {
+          b0 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 16 * (lane >> 5));
+          b1 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+        }
         dkacc[cb] = mfma_32x32x16_bf16(df0, b0, dkacc[cb]);
         dkacc[cb] = mfma_32x32x16_bf16(df1, b1, dkacc[cb]);
       }

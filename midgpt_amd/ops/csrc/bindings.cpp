@@ -266,6 +266,27 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
                          (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),    \
                          lse.data_ptr<float>(), delta.data_ptr<float>(),        \
                          (u16*)dk.data_ptr(), (u16*)dv.data_ptr(), B, H, T);    \
+    else if (abl && abl[0] == '3')                                              \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NA, 3>), dim3(grid_a),        \
+                         dim3(NA * 64), smem_a, cur_stream(),                   \
+                         (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),   \
+                         (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),    \
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),        \
+                         (u16*)dk.data_ptr(), (u16*)dv.data_ptr(), B, H, T);    \
+    else if (abl && abl[0] == '4')                                              \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NA, 4>), dim3(grid_a),        \
+                         dim3(NA * 64), smem_a, cur_stream(),                   \
+                         (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),   \
+                         (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),    \
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),        \
+                         (u16*)dk.data_ptr(), (u16*)dv.data_ptr(), B, H, T);    \
+    else if (abl && abl[0] == '5')                                              \
+      hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NA, 5>), dim3(grid_a),        \
+                         dim3(NA * 64), smem_a, cur_stream(),                   \
+                         (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),   \
+                         (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),    \
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),        \
+                         (u16*)dk.data_ptr(), (u16*)dv.data_ptr(), B, H, T);    \
     else if (abl && abl[0] == '2')                                              \
       hipLaunchKernelGGL((attn_bwd_dkv_kernel<CC, NA, 2>), dim3(grid_a),        \
                          dim3(NA * 64), smem_a, cur_stream(),                   \
