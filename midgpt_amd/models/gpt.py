@@ -160,7 +160,6 @@ class GPT(nn.Module):
 
     def forward(self, idx: torch.Tensor) -> torch.Tensor:
         """idx (B, T) int64 -> logits (B, T, V) in compute dtype."""
-        dtype = self.wte.dtype
         x = self.drop(F.embedding(idx, self.wte))
         for blk in self.blocks:
             if self.remat and torch.is_grad_enabled():
