@@ -59,8 +59,11 @@ def main():
     config.g_accum_iters = g_accum
     # ZeRO sharding active whenever world > 1 (headline config shard_model=True)
 
-    torch.manual_seed(1234 + rank)
+    # identical seed for model init on EVERY rank (data-parallel replicas
+    # must start from the same weights); per-rank seed afterwards for data
+    torch.manual_seed(1234)
     model, engine = build_engine(config, device)
+    torch.manual_seed(1234 + rank)
     tokens_per_step_per_gpu = args.local_batch * mc.block_size
 
     # pre-generate a couple of synthetic batches on device (data pipeline is

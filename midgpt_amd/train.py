@@ -63,10 +63,15 @@ def evaluate(model: GPT, loader: BatchLoader, batch_size: int,
 
 def train(config: ExperimentConfig):
     rank, world, device = pdist.init_distributed()
-    torch.manual_seed(1234 + rank if config.seed is None else config.seed + rank)
-    gen = torch.Generator().manual_seed(config.seed) if config.seed is not None else None
+    # identical seed on every rank for MODEL INIT (data-parallel replicas
+    # must start from the same weights); re-seed per rank below for
+    # data/dropout streams.
+    base_seed = 1234 if config.seed is None else config.seed
+    torch.manual_seed(base_seed)
+    gen = torch.Generator().manual_seed(base_seed)
 
     model, engine = build_engine(config, device, gen)
+    torch.manual_seed(base_seed + rank)
     if pdist.is_main():
         print(f"params: {count_params(model)/1e6:.1f}M  world={world} "
               f"device={device} zero={engine.zero}")
