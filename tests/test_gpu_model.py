@@ -198,3 +198,31 @@ def test_two_process_one_gpu_zero_engine(tmp_path):
     for r, full in res.items():
         rel = (full - ref[:full.numel()]).norm() / ref[:full.numel()].norm()
         assert rel < 5e-3, (r, float(rel))
+
+
+def test_training_trajectory_tracks_cpu_reference():
+    """20 steps of the tiny model: the GPU bf16 HIP path's loss curve must
+    track the CPU fp32 reference path (systematic kernel-numerics drift
+    beyond per-op tolerances would separate the curves)."""
+    def run(device, dtype):
+        torch.manual_seed(0)
+        model = GPT(SMALL).to(device)
+        engine = ShardedAdamW(model, compute_dtype=dtype, zero=False,
+                              peak_lr=1e-2)
+        g = torch.Generator().manual_seed(1)
+        x, y = synthetic_batch(512, 128, 8, 1, generator=g)
+        xs, ys = x[0].to(device), y[0].to(device)
+        losses = []
+        for _ in range(20):
+            loss = model.loss(xs, ys)
+            loss.backward()
+            engine.microstep_end()
+            engine.step(3e-3)
+            losses.append(float(loss.detach()))
+        return losses
+
+    gpu = run("cuda", torch.bfloat16)
+    cpu = run("cpu", torch.float32)
+    diffs = [abs(a - b) for a, b in zip(gpu, cpu)]
+    assert max(diffs[:10]) < 0.08, (gpu[:10], cpu[:10])
+    assert abs(gpu[-1] - cpu[-1]) < 0.3, (gpu[-1], cpu[-1])
