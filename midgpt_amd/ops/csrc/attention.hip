@@ -8,11 +8,15 @@
 // (reference src/model.py:71-79). The T x T score matrix is never
 // materialized.
 //
-// Geometry (fwd): one workgroup = 4 waves = 128 q rows (32/wave);
-// KV tiles of 32 rows double-buffered in LDS; grid = B*H*(T/128).
-// Geometry (bwd): one workgroup = 4 waves = 128 k rows; iterates q tiles;
-// dK/dV accumulate in registers, dQ via fp32 atomics (v1; split-q kernel
-// is the planned v2).
+// Geometry (fwd): one workgroup = NW waves (8 at T%256==0, else 4), each
+// wave owning 32 q rows; KV tiles of 32 rows double-buffered in LDS with
+// T14 load-early/write-late staging; grid = B*H*(T/(NW*32)), ordered so
+// one (b,h)'s q-blocks share an XCD's L2.
+// Geometry (bwd): TWO kernels — dkv (workgroup = NW*32 k rows, iterates
+// q tiles >= its diagonal; dK/dV accumulate in registers) and dq
+// (workgroup = NW*32 q rows, iterates k tiles; dQ in registers, no
+// atomics), both with double-buffered T14 staging. A delta kernel
+// (rowsum(dO*O)) runs first.
 #include "common.h"
 #include "mfma.h"
 
