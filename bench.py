@@ -39,6 +39,8 @@ def main():
                    help="per-block activation recompute (default OFF for the "
                         "bench: 288 GB HBM fits stored activations at T=1024, "
                         "saving the recompute forward)")
+    p.add_argument("--no-remat", action="store_true",
+                   help="force remat OFF even at seq > 1024 (memory permitting)")
     args = p.parse_args()
 
     rank, world, device = pdist.init_distributed()
@@ -49,7 +51,8 @@ def main():
     config = load_config(args.config)
     config.synthetic_data = True
     config.rundir = ""
-    config.remat = args.remat or config.model_config.block_size > 1024
+    config.remat = (args.remat or config.model_config.block_size > 1024) \
+        and not args.no_remat
     mc = config.model_config
     micro = args.micro_batch or min(args.local_batch,
                                     64 if mc.block_size <= 1024 else 8)
