@@ -16,7 +16,6 @@ import time
 
 import torch
 
-from midgpt_amd import ops
 from midgpt_amd.config import ExperimentConfig
 from midgpt_amd.data import BatchLoader
 from midgpt_amd.models.gpt import GPT, count_params
