@@ -19,7 +19,7 @@ batches = [synthetic_batch(mc.vocab_size, mc.block_size, 32, 1, device="cuda",
                            generator=g) for _ in range(4)]
 t0 = time.perf_counter()
 mem0 = None
-for it in range(300):
+for it in range(1000):
     lr = warmup_cosine_lr(it, peak_lr=1e-3, warmup_steps=100,
                           decay_steps=25000, min_lr=1e-5)
     x, y = batches[it % 4]
@@ -30,7 +30,7 @@ for it in range(300):
     if it == 20:
         torch.cuda.synchronize()
         mem0 = torch.cuda.memory_allocated()
-    if it % 100 == 99:
+    if it % 250 == 249:
         torch.cuda.synchronize()
         lv = float(loss.detach())
         mem = torch.cuda.memory_allocated()
