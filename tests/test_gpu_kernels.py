@@ -95,7 +95,8 @@ def test_qkv_prep_fwd_bwd(B, T, H, C):
 
 
 @pytest.mark.parametrize("B,H,T,C", [(2, 2, 128, 64), (1, 2, 256, 128),
-                                     (2, 1, 1024, 128), (2, 2, 1024, 64)])
+                                     (2, 1, 1024, 128), (2, 2, 1024, 64),
+                                     (1, 1, 4096, 128), (1, 1, 2048, 64)])
 def test_attention_fwd(B, H, T, C):
     torch.manual_seed(4)
     q = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16)
@@ -113,7 +114,8 @@ def test_attention_fwd(B, H, T, C):
 
 
 @pytest.mark.parametrize("B,H,T,C", [(2, 2, 128, 64), (1, 2, 256, 128),
-                                     (1, 1, 1024, 64), (1, 1, 1024, 128)])
+                                     (1, 1, 1024, 64), (1, 1, 1024, 128),
+                                     (1, 1, 4096, 128)])
 def test_attention_bwd(B, H, T, C):
     torch.manual_seed(5)
     q = torch.randn(B, H, T, C, device=DEV, dtype=torch.bfloat16, requires_grad=True)
