@@ -197,6 +197,7 @@ class _Attention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v):
         if _use_hip(q):
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
             o, lse = _C.attn_fwd(q, k, v)
         else:
             B, H, T, C = q.shape
