@@ -73,5 +73,13 @@ class ExperimentConfig:
 def load_config(name: str) -> ExperimentConfig:
     """Dynamic config import by module name, mirroring launch.py:25-27."""
     import importlib
-    mod = importlib.import_module(f"midgpt_amd.configs.{name}")
+    try:
+        mod = importlib.import_module(f"midgpt_amd.configs.{name}")
+    except ModuleNotFoundError as e:
+        import pkgutil
+        import midgpt_amd.configs as cfgs
+        avail = sorted(m.name for m in pkgutil.iter_modules(cfgs.__path__)
+                       if not m.name.startswith("_"))
+        raise SystemExit(f"unknown config '{name}' ({e}); available: "
+                         f"{', '.join(avail)}")
     return mod.config
