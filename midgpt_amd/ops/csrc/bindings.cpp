@@ -192,13 +192,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   TORCH_CHECK(C == 64 || C == 128, "attn: head dim 64 or 128 (got ", C, ")");
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, T}, q.options().dtype(torch::kFloat));
-  // Mirrored-strip scheduling (SPW=2): causal work per wave is equal, so
-  // staging barriers park nobody. C=128 runs 4-wave WGs at 1 wave/SIMD
-  // (whole 512-register file: 256 VGPR + ~190 AGPR accumulators);
-  // C=64 fits 8-wave WGs at 2 waves/SIMD. Fallbacks for small T.
-  // Measured best: 8-wave WGs at 2 waves/SIMD (WG co-residency beats both
-  // 4-wave low-bandwidth and the 1-wave/SIMD mirrored-strip variant —
-  // SQ_WAIT is memory-wait dominated, so TLP matters most).
+  // Measured best: 8-wave WGs at 2 waves/SIMD, SPW=1. The SPW=2
+  // mirrored-strip variant (1 wave/SIMD for C=128) and 3-waves/SIMD were
+  // both measured slower — SQ_WAIT is memory-wait dominated, so wave
+  // co-residency (TLP) matters most. Fallback to 4 waves for small T.
   const int NW = (T % 256 == 0) ? 8 : 4;
   const int SPW = 1;
   long grid = (long)B * H * (T / (NW * 32 * SPW));
