@@ -2,7 +2,7 @@
 reference implementations on CPU.
 
 The HIP extension is built IN-TREE as ``midgpt_amd/ops/_C*.so`` (see
-``midgpt_amd/ops/build.py`` / ``__graft_entry__.build``). On a GPU box the
+``setup.py`` / ``__graft_entry__.build``). On a GPU box the
 extension is REQUIRED: ops raise if it is missing so that a silent eager
 fallback can never masquerade as the native path. Set ``MIDGPT_FORCE_REF=1``
 to explicitly run the reference path on GPU (debugging only).
