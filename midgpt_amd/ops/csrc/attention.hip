@@ -310,6 +310,189 @@ __global__ __launch_bounds__(NW * 64, MINW) void attn_fwd_kernel(const u16* __re
 }
 
 // ===========================================================================
+// Forward, PAIRED variant: two 32-row KV tiles per barrier round with ONE
+// merged online-softmax rescale — halves barrier rounds and O-rescale
+// passes vs the single-tile loop. nkt is always even (multiple of NW).
+// ===========================================================================
+template <int C, int NW>
+__global__ __launch_bounds__(NW * 64, 2) void attn_fwd2_kernel(
+    const u16* __restrict__ q, const u16* __restrict__ k,
+    const u16* __restrict__ v, u16* __restrict__ o, float* __restrict__ lse,
+    int B, int H, int T) {
+  constexpr int NCB = C / 32;
+  constexpr int NCH = C / 16;
+  const float scale = rsqrtf((float)C);
+  const long bh = blockIdx.x % ((long)B * H);
+  const int qb = blockIdx.x / (B * H);
+  const int q0 = qb * (NW * 32);
+  const int lane = lane_id();
+  const int w = wave_id();
+  const int qw0 = q0 + 32 * w;
+  const int myq = qw0 + (lane & 31);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* ldsK = (u16*)smem;                       // [2][2][32*C]
+  u16* ldsVt = (u16*)(smem + 2 * 64 * C * 2);   // [2][2][C*32]
+  float* obuf = (float*)smem;                   // epilogue reuse
+
+  const u16* qg = q + (bh * T) * C;
+  const u16* kg = k + (bh * T) * C;
+  const u16* vg = v + (bh * T) * C;
+
+  bf16x8_t qf[NCH];
+  {
+    const u16* qrow = qg + (long)myq * C;
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      qf[ch] = *(const bf16x8_t*)(qrow + 16 * ch + 8 * (lane >> 5));
+  }
+  f32x16 oacc[NCB];
+#pragma unroll
+  for (int cb = 0; cb < NCB; ++cb) oacc[cb] = (f32x16)(0.f);
+  float m = -1e30f, lsum = 0.f;
+
+  const int nrounds = (q0 + NW * 32) / 64;  // nkt/2, nkt always even
+  // prologue: stage round 0 (two tiles) synchronously
+  stage_rm<C, NW * 64>(kg, ldsK);
+  stage_rm<C, NW * 64>(kg + 32 * C, ldsK + 32 * C);
+  stage_tr<C, NW * 64>(vg, ldsVt);
+  stage_tr<C, NW * 64>(vg + 32 * C, ldsVt + C * 32);
+  __syncthreads();
+
+  RmStage<C, NW * 64> kstA, kstB;
+  TrStage<C, NW * 64> vstA, vstB;
+  for (int rd = 0; rd < nrounds; ++rd) {
+    const int buf = rd & 1;
+    const bool pre = rd + 1 < nrounds;
+    if (pre) {
+      const long nb = (long)(rd + 1) * 64;
+      kstA.load(kg + nb * C);
+      kstB.load(kg + (nb + 32) * C);
+      vstA.load(vg + nb * C);
+      vstB.load(vg + (nb + 32) * C);
+    }
+    const int k0 = rd * 64;
+    if (k0 <= qw0 + 31) {  // at least sub-tile A needed by this wave
+      const u16* kbuf = ldsK + buf * 64 * C;
+      const u16* vbuf = ldsVt + buf * 64 * C;  // (C*32 per sub-tile) x 2
+      const bool needB = (k0 + 32) <= qw0 + 31;
+      // S for both sub-tiles (independent MFMA chains)
+      f32x16 s0 = (f32x16)(0.f), s1 = (f32x16)(0.f);
+#pragma unroll
+      for (int ch = 0; ch < NCH; ++ch) {
+        bf16x8_t a0 = read_rm_frag<C>(kbuf, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+        s0 = mfma_32x32x16_bf16(a0, qf[ch], s0);
+      }
+      if (needB) {
+#pragma unroll
+        for (int ch = 0; ch < NCH; ++ch) {
+          bf16x8_t a1 = read_rm_frag<C>(kbuf + 32 * C, lane & 31,
+                                        16 * ch * 2 + 16 * (lane >> 5));
+          s1 = mfma_32x32x16_bf16(a1, qf[ch], s1);
+        }
+      }
+      // merged mask + stats over up to 64 k
+      float sv0[16], sv1[16];
+      float mt = -1e30f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        sv0[r] = (k0 + mfma_d_row(lane, r) > myq) ? -1e30f : s0[r];
+        mt = fmaxf(mt, sv0[r]);
+        sv1[r] = (!needB || k0 + 32 + mfma_d_row(lane, r) > myq) ? -1e30f : s1[r];
+        mt = fmaxf(mt, sv1[r]);
+      }
+      mt = fmaxf(mt, __shfl_xor(mt, 32));
+      const float mn = fmaxf(m, mt);
+      const float alpha = __expf((m - mn) * scale);
+      const bool need_rescale = !__all(mt <= m);
+      m = mn;
+      float p0[16], p1[16], psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        p0[r] = __expf((sv0[r] - mn) * scale);
+        p1[r] = __expf((sv1[r] - mn) * scale);
+        psum += p0[r] + p1[r];
+      }
+      psum += __shfl_xor(psum, 32);
+      lsum = lsum * alpha + psum;
+      if (need_rescale) {  // ONE rescale per 64 k
+        float arow[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) arow[r] = shfl32(alpha, mfma_d_row(lane, r));
+#pragma unroll
+        for (int cb = 0; cb < NCB; ++cb)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) oacc[cb][r] *= arow[r];
+      }
+      // PV for both sub-tiles
+      bf16x8_t pA0 = dlayout_to_afrag(p0);
+      bf16x8_t pA1 = dlayout_to_afrag(p0 + 8);
+#pragma unroll
+      for (int cb = 0; cb < NCB; ++cb) {
+        bf16x8_t b0 = read_tr_frag(vbuf, 32 * cb + (lane & 31), 16 * (lane >> 5));
+        bf16x8_t b1 = read_tr_frag(vbuf, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+        oacc[cb] = mfma_32x32x16_bf16(pA0, b0, oacc[cb]);
+        oacc[cb] = mfma_32x32x16_bf16(pA1, b1, oacc[cb]);
+      }
+      if (needB) {
+        bf16x8_t pB0 = dlayout_to_afrag(p1);
+        bf16x8_t pB1 = dlayout_to_afrag(p1 + 8);
+        const u16* vbufB = vbuf + C * 32;
+#pragma unroll
+        for (int cb = 0; cb < NCB; ++cb) {
+          bf16x8_t b0 = read_tr_frag(vbufB, 32 * cb + (lane & 31), 16 * (lane >> 5));
+          bf16x8_t b1 = read_tr_frag(vbufB, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+          oacc[cb] = mfma_32x32x16_bf16(pB0, b0, oacc[cb]);
+          oacc[cb] = mfma_32x32x16_bf16(pB1, b1, oacc[cb]);
+        }
+      }
+    }
+    if (pre) {
+      u16* kd = ldsK + (1 - buf) * 64 * C;
+      u16* vd = ldsVt + (1 - buf) * 64 * C;
+      kstA.write(kd);
+      kstB.write(kd + 32 * C);
+      vstA.write_tr(vd);
+      vstB.write_tr(vd + C * 32);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: identical to the single-tile kernel
+  const float rec = 1.f / lsum;
+  float rrow[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) rrow[r] = shfl32(rec, mfma_d_row(lane, r));
+  if (lane < 32) lse[bh * T + myq] = m * scale + __logf(lsum);
+  float* ob = obuf + w * 32 * 32;
+  u16* og = o + (bh * T + qw0) * C;
+#pragma unroll
+  for (int cb = 0; cb < NCB; ++cb) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = mfma_d_row(lane, r);
+      *(float*)((char*)ob + row * 128 + (((lane & 31) * 4) ^ ((row & 7) << 4))) =
+          oacc[cb][r] * rrow[r];
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    const int row = lane & 31;
+    const int c16 = 16 * (lane >> 5);
+    float tmp[16];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      f32x4 t = *(const f32x4*)((char*)ob + row * 128 + (((c16 + 4 * i) * 4) ^ ((row & 7) << 4)));
+      tmp[4 * i] = t[0]; tmp[4 * i + 1] = t[1]; tmp[4 * i + 2] = t[2]; tmp[4 * i + 3] = t[3];
+    }
+    u16x8 out0, out1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { out0[j] = f2b(tmp[j]); out1[j] = f2b(tmp[8 + j]); }
+    *(u16x8*)(og + (long)row * C + 32 * cb + c16) = out0;
+    *(u16x8*)(og + (long)row * C + 32 * cb + c16 + 8) = out1;
+    __builtin_amdgcn_s_waitcnt(0);
+  }
+}
+
+// ===========================================================================
 // delta = rowsum(dO * O) — one wave per row (prologue of backward)
 // ===========================================================================
 __global__ void attn_delta_kernel(const u16* __restrict__ dO,

@@ -206,6 +206,30 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
                      smem, cur_stream(), (const u16*)q.data_ptr(),              \
                      (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),        \
                      (u16*)o.data_ptr(), lse.data_ptr<float>(), B, H, T)
+  // paired-tile kernel (one merged rescale per 64 k): measured +4% at
+  // C=128, -4% at C=64 -> default for C=128 only (MIDGPT_ATTN_FWD2=1
+  // forces it everywhere, MIDGPT_ATTN_FWD1=1 disables).
+  static const bool fwd2_force = getenv("MIDGPT_ATTN_FWD2") != nullptr;
+  static const bool fwd1_force = getenv("MIDGPT_ATTN_FWD1") != nullptr;
+  const bool fwd2 = !fwd1_force && (fwd2_force || C == 128);
+  if (fwd2 && NW == 8) {
+    size_t smem2 = std::max((size_t)(8 * 32 * C * 2), (size_t)(NW * 32 * 32 * 4));
+#define LAUNCH_FWD2(CC)                                                         \
+    do {                                                                        \
+      if (smem2 > 64 * 1024)                                                    \
+        hipFuncSetAttribute(                                                    \
+            reinterpret_cast<const void*>(&attn_fwd2_kernel<CC, 8>),            \
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem2);            \
+      hipLaunchKernelGGL((attn_fwd2_kernel<CC, 8>), dim3(grid), dim3(512),      \
+                         smem2, cur_stream(), (const u16*)q.data_ptr(),         \
+                         (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),    \
+                         (u16*)o.data_ptr(), lse.data_ptr<float>(), B, H, T);   \
+    } while (0)
+    if (C == 128) LAUNCH_FWD2(128); else LAUNCH_FWD2(64);
+#undef LAUNCH_FWD2
+    launch_check();
+    return {o, lse};
+  }
   static const bool minw3 = getenv("MIDGPT_ATTN_FWD_MINW3") != nullptr;
   if (C == 128 && NW == 8 && minw3) LAUNCH_FWD(128, 8, 1, 3);
   else if (C == 128 && NW == 8) LAUNCH_FWD(128, 8, 1, 2);
