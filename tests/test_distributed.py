@@ -116,3 +116,50 @@ def test_gloo_collective_helpers(tmp_path):
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
+
+
+def _odd_worker(rank, world, init_file, out_q):
+    """ZeRO with a parameter count NOT divisible by world — exercises the
+    padded shard tail (the 1.5B/8-GPU case)."""
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    import torch.nn as nn
+    from midgpt_amd.parallel.engine import ShardedAdamW
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Linear(3, 5, bias=True), nn.Linear(5, 3, bias=False))
+    # total params = 15 + 5 + 15 = 35, odd -> padded to 36 at world 2
+    engine = ShardedAdamW(model, compute_dtype=torch.float32, zero=True)
+    assert engine.padded == 36 and engine.shard_size == 18
+    g = torch.Generator().manual_seed(1)
+    x = torch.randn(8, 3, generator=g)
+    y = torch.randn(8, 3, generator=g)
+    for _ in range(3):
+        ((model(x[rank * 4:(rank + 1) * 4]) -
+          y[rank * 4:(rank + 1) * 4]) ** 2).mean().backward()
+        engine.microstep_end()
+        engine.step(1e-2)
+    full = torch.zeros(engine.padded)
+    full[engine.shard_off:engine.shard_off + engine.shard_size] = engine.master
+    dist.all_reduce(full)
+    out_q.put((rank, full[:engine.total].numpy().copy()))
+    dist.destroy_process_group()
+
+
+def test_zero_padded_shard_tail(tmp_path):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    init_file = str(tmp_path / "pg_odd")
+    procs = [ctx.Process(target=_odd_worker, args=(r, 2, init_file, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(2):
+        r, full = q.get()
+        res[r] = torch.from_numpy(full)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    # both ranks assemble the same full master; finite and moved from init
+    assert torch.allclose(res[0], res[1])
+    assert torch.isfinite(res[0]).all()
