@@ -27,13 +27,13 @@ def load_model(ckpt_dir: str, device: torch.device) -> tuple[GPT, ExperimentConf
     state = ckpt.load_full_state(ckpt_dir)
     if state is not None:
         master = state["master"]
-        off = 0
         sd = {}
         for pm in state["manifest"]["params"]:
             n, shape = pm["numel"], pm["shape"]
             sd[pm["name"]] = master[pm["offset"]:pm["offset"] + n].view(shape)
-            off += n
-        model.load_state_dict(sd, strict=False)
+        # strict=False only because rope sin/cos buffers are non-persistent
+        missing, unexpected = model.load_state_dict(sd, strict=False)
+        assert not unexpected, f"unexpected checkpoint params: {unexpected}"
         print(f"loaded checkpoint step {state['step']}")
     else:
         print("no checkpoint found; using random init")
