@@ -25,6 +25,22 @@ from midgpt_amd.train import build_engine
 from midgpt_amd.utils.lr import warmup_cosine_lr
 
 
+def _self_launch(n: int):
+    """Re-exec under torch.distributed.run with one rank per GPU.
+
+    The driver may invoke ``python bench.py --gpus 8`` directly; without
+    this, WORLD_SIZE is unset and the run would silently measure 1 GPU.
+    """
+    import subprocess
+    import sys
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--nnodes=1", f"--nproc-per-node={n}",
+           "--master-addr=127.0.0.1", "--master-port=29771",
+           sys.argv[0], *sys.argv[1:]]
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    raise SystemExit(subprocess.call(cmd))
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -43,9 +59,11 @@ def main():
                    help="force remat OFF even at seq > 1024 (memory permitting)")
     args = p.parse_args()
 
+    if args.gpus > 1 and int(os.environ.get("WORLD_SIZE", "1")) <= 1:
+        _self_launch(args.gpus)  # re-exec under torchrun; does not return
+
     rank, world, device = pdist.init_distributed()
-    assert world == args.gpus or world == 1, \
-        f"WORLD_SIZE={world} != --gpus {args.gpus}"
+    assert world == args.gpus, f"WORLD_SIZE={world} != --gpus {args.gpus}"
     n = max(world, 1)
 
     config = load_config(args.config)
@@ -115,11 +133,13 @@ def main():
 
     ms_per_step = elapsed / args.steps * 1000.0
     tokens_per_sec = n * tokens_per_step_per_gpu * args.steps / elapsed
-    # model FLOPs/token = 6*N_params + attention term (12*L*D*T per token
-    # ~= 6*L*D*T*2? use 6N + 12*L*T*D); MFU vs 2.5 PF dense bf16 per GPU
+    # model FLOPs/token = 6*(matmul params incl. the UNTIED lm_head, which
+    # is compute-bearing even though count_params excludes it for reference
+    # parity) + 12*L*D*T attention term; MFU vs 2.5 PF dense bf16 per GPU.
     from midgpt_amd.models.gpt import count_params
     n_params = count_params(model)
-    flops_per_token = 6 * n_params + 12 * mc.n_layer * mc.n_embd * mc.block_size
+    n_matmul = n_params + model.lm_head.weight.numel()
+    flops_per_token = 6 * n_matmul + 12 * mc.n_layer * mc.n_embd * mc.block_size
     mfu = tokens_per_sec * flops_per_token / (n * 2.5e15)
 
     if rank == 0:

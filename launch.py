@@ -31,10 +31,23 @@ def main():
     config = load_config(args.config)
     config.debug = args.debug
 
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if args.distributed and world <= 1:
+        p.error("--distributed requires a torchrun launch (WORLD_SIZE unset); "
+                "use: python -m torch.distributed.run --nproc-per-node N "
+                "--master-addr 127.0.0.1 launch.py ...")
+
     rundir = args.rundir
     if rundir is None:
+        # Every rank must agree on the generated rundir, or checkpoint
+        # shards scatter across per-rank timestamp dirs (reference
+        # launch.py asserts multihost runs prespecify rundir): broadcast
+        # rank 0's generated name over the process group.
         stamp = datetime.datetime.now().strftime("%Y%m%d_%H%M%S")
         rundir = os.path.join("runs", f"{args.config}_{stamp}")
+        if world > 1:
+            pdist.init_distributed()
+            rundir = pdist.broadcast_str(rundir)
     config.rundir = rundir
 
     # rank 0 creates the rundir and freezes the config

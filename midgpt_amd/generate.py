@@ -73,7 +73,9 @@ def generate(model, idx: torch.Tensor, max_new_tokens: int,
     model.eval()
     block = model.config.block_size
     cache = [None] * model.config.n_layer
-    logits = _forward_cached(model, idx, cache, 0)
+    # prompts longer than block_size condition on the last block_size tokens
+    # (reference sample.py:74-81 crops to the trailing window)
+    logits = _forward_cached(model, idx[:, -block:], cache, 0)
     for _ in range(max_new_tokens):
         nxt = _sample(logits, temperature, top_k, generator)
         idx = torch.cat([idx, nxt[:, None]], dim=1)

@@ -62,6 +62,22 @@ def _is_gloo() -> bool:
     return dist.get_backend() == "gloo"
 
 
+def broadcast_str(s: str, src: int = 0, max_len: int = 1024) -> str:
+    """Broadcast a short string from ``src`` to all ranks (used for the
+    generated rundir name; reference multihost runs prespecify it)."""
+    if not dist.is_initialized():
+        return s
+    buf = torch.zeros(max_len, dtype=torch.uint8)
+    if get_rank() == src:
+        raw = s.encode()[:max_len]
+        buf[:len(raw)] = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+    if torch.cuda.is_available() and dist.get_backend() == "nccl":
+        buf = buf.cuda()
+    dist.broadcast(buf, src=src)
+    raw = bytes(buf.cpu().tolist())
+    return raw.rstrip(b"\x00").decode()
+
+
 def all_reduce_(t: torch.Tensor, op=dist.ReduceOp.SUM if dist.is_available() else None):
     if dist.is_initialized():
         dist.all_reduce(t, op=op)

@@ -47,17 +47,26 @@ def build_engine(config: ExperimentConfig, device: torch.device,
 
 
 @torch.no_grad()
-def evaluate(model: GPT, loader: BatchLoader, batch_size: int,
+def evaluate(model: GPT, loader: BatchLoader, split: str, batch_size: int,
              device: torch.device, n_batches: int = 200) -> float:
-    """Mean loss over fixed-size batches (reference src/train.py:107-117)."""
+    """Mean loss on ``split`` over fixed-size batches, all-reduced (mean)
+    across ranks so every rank sees the global estimate (reference
+    src/train.py:107-117,195-196 evaluates train AND val data)."""
     model.eval()
     total = 0.0
     for _ in range(n_batches):
-        x, y = loader.batch("val", batch_size, 1)
+        x, y = loader.batch(split, batch_size, 1)
         x, y = x[0].to(device), y[0].to(device)
         total += float(model.loss(x, y))
     model.train()
-    return total / n_batches
+    mean = total / n_batches
+    world = pdist.get_world_size()
+    if world > 1:
+        t = torch.tensor([mean], dtype=torch.float64,
+                         device=device if device.type == "cuda" else "cpu")
+        pdist.all_reduce_(t)
+        mean = float(t[0]) / world
+    return mean
 
 
 def train(config: ExperimentConfig):
@@ -107,8 +116,8 @@ def train(config: ExperimentConfig):
             pbar = None
     for it in range(first_step, config.max_steps):
         if it % config.eval_interval == 0:
-            tl = evaluate(model, loader, local_bs, device, eval_batches)
-            vl = evaluate(model, loader, local_bs, device, eval_batches)
+            tl = evaluate(model, loader, "train", local_bs, device, eval_batches)
+            vl = evaluate(model, loader, "val", local_bs, device, eval_batches)
             if pdist.is_main():
                 print(f"step {it}: loss/train {tl:.4f} loss/val {vl:.4f}")
             log_metrics(config, it, {"loss/train": tl, "loss/val": vl})

@@ -207,6 +207,48 @@ def test_checkpoint_max_to_keep(tmp_path):
     assert dirs == ["ckpt_0000002"]
 
 
+def test_generate_prompt_longer_than_block():
+    """A prompt with T0 > block_size must condition on its trailing window
+    (reference sample.py crops to block_size) instead of raising."""
+    torch.manual_seed(12)
+    from midgpt_amd.generate import generate
+    model = GPT(TINY)  # block_size 16
+    idx = torch.randint(0, 37, (1, 25))
+    out = generate(model, idx, 5, temperature=0.0)
+    assert out.shape == (1, 30)
+    assert torch.equal(out[:, :25], idx)
+    cur = idx.clone()
+    for _ in range(5):
+        logits = model(cur[:, -16:])
+        cur = torch.cat([cur, logits[:, -1].argmax(-1)[:, None]], dim=1)
+    assert torch.equal(out, cur)
+
+
+def test_evaluate_reads_requested_split(tmp_path):
+    """loss/train must come from train.bin, loss/val from val.bin
+    (reference src/train.py:195-196 evaluates both datasets)."""
+    from midgpt_amd.train import evaluate
+    # distinct constant token streams per split
+    np.full(4000, 3, dtype=np.uint16).tofile(tmp_path / "train.bin")
+    np.full(4000, 11, dtype=np.uint16).tofile(tmp_path / "val.bin")
+    bl = BatchLoader(str(tmp_path), 37, 16, seed=0)
+
+    class TokenMeanModel:
+        def eval(self):
+            pass
+
+        def train(self):
+            pass
+
+        def loss(self, x, y):
+            return x.float().mean()
+
+    m = TokenMeanModel()
+    dev = torch.device("cpu")
+    assert evaluate(m, bl, "train", 2, dev, n_batches=3) == pytest.approx(3.0)
+    assert evaluate(m, bl, "val", 2, dev, n_batches=3) == pytest.approx(11.0)
+
+
 def test_train_entrypoint_runs(tmp_path):
     from midgpt_amd.train import train
     cfg = tiny_config(tmp_path, max_steps=3, eval_interval=2, debug=True)
