@@ -160,7 +160,7 @@ class GPT(nn.Module):
 
     def forward(self, idx: torch.Tensor) -> torch.Tensor:
         """idx (B, T) int64 -> logits (B, T, V) in compute dtype."""
-        x = self.drop(F.embedding(idx, self.wte))
+        x = self.drop(ops.embedding(idx, self.wte))
         for blk in self.blocks:
             if self.remat and torch.is_grad_enabled():
                 x = torch.utils.checkpoint.checkpoint(

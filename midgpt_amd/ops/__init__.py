@@ -239,6 +239,38 @@ def flash_attention(q, k, v):
 
 
 # ----------------------------------------------------------------------------
+# Embedding: gather fwd, scatter-add bwd (reference src/layers.py:13-34;
+# plan K8). The forward gather is a plain coalesced index_select (torch);
+# the backward is the hand-written HIP fp32-accurate scatter-add.
+# ----------------------------------------------------------------------------
+class _Embedding(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, idx, weight):
+        ctx.save_for_backward(idx)
+        ctx.V = weight.shape[0]
+        ctx.wdtype = weight.dtype
+        return weight[idx]
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        D = dy.shape[-1]
+        dy2d = dy.reshape(-1, D)
+        if _use_hip(dy) and dy.dtype == torch.bfloat16:
+            dw = _C.embedding_bwd(dy2d.contiguous(), idx.reshape(-1), ctx.V)
+        else:
+            dw32 = torch.zeros(ctx.V, D, dtype=torch.float32, device=dy.device)
+            dw32.index_add_(0, idx.reshape(-1), dy2d.float())
+            dw = dw32.to(ctx.wdtype)
+        return None, dw
+
+
+def embedding(idx, weight):
+    """idx (...,) int64, weight (V, D) -> (..., D)."""
+    return _Embedding.apply(idx, weight)
+
+
+# ----------------------------------------------------------------------------
 # Fused softmax cross-entropy over the vocab (reference src/train.py:76-77;
 # plan K9). Never materializes the fp32 softmax over V on the HIP path.
 # ----------------------------------------------------------------------------

@@ -339,6 +339,29 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
 }
 
 // ---------------------------------------------------------------------------
+torch::Tensor embedding_bwd(torch::Tensor dy, torch::Tensor idx, long V) {
+  CHECK_GPU(dy); CHECK_GPU(idx);
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16, "embedding_bwd: dy bf16");
+  TORCH_CHECK(idx.scalar_type() == torch::kLong, "embedding_bwd: idx int64");
+  long N = dy.size(0);
+  int D = dy.size(1);
+  TORCH_CHECK(D % 8 == 0, "embedding_bwd: D % 8 == 0");
+  auto dw32 = torch::zeros({V, (long)D}, dy.options().dtype(torch::kFloat));
+  long grid = std::min(N, (long)4096);
+  hipLaunchKernelGGL(embed_bwd_scatter_kernel, dim3(grid), dim3(256), 0,
+                     cur_stream(), (const u16*)dy.data_ptr(),
+                     idx.data_ptr<long>(), dw32.data_ptr<float>(), N, D);
+  launch_check();
+  auto dw = torch::empty({V, (long)D}, dy.options());
+  long n = V * (long)D;
+  long cgrid = std::min((n / 4 + 255) / 256 + 1, (long)4096);
+  hipLaunchKernelGGL(f32_to_bf16_kernel, dim3(cgrid), dim3(256), 0,
+                     cur_stream(), dw32.data_ptr<float>(), (u16*)dw.data_ptr(), n);
+  launch_check();
+  return dw;
+}
+
+// ---------------------------------------------------------------------------
 torch::Tensor probe_mfma(torch::Tensor A, torch::Tensor B) {
   CHECK_GPU(A); CHECK_GPU(B);
   auto D = torch::zeros({32, 32}, A.options());
@@ -368,6 +391,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adamw_step", &adamw_step);
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_bwd", &attn_bwd);
+  mod.def("embedding_bwd", &embedding_bwd);
   mod.def("linear_gelu_fwd", &linear_gelu_fwd);
   mod.def("lt_probe", &lt_probe);
   mod.def("matmul_dgelu", &matmul_dgelu);

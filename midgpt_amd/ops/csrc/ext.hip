@@ -5,5 +5,6 @@
 #include "ce.hip"
 #include "adamw.hip"
 #include "attention.hip"
+#include "embedding.hip"
 #include "probe.hip"
 #include "bindings.cpp"
