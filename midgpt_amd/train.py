@@ -23,6 +23,7 @@ from midgpt_amd.parallel import dist as pdist
 from midgpt_amd.parallel.engine import ShardedAdamW
 from midgpt_amd.utils import checkpoint as ckpt
 from midgpt_amd.utils.lr import warmup_cosine_lr
+from midgpt_amd.utils.prefetch import DevicePrefetcher
 
 DTYPES = {"float32": torch.float32, "bfloat16": torch.bfloat16}
 
@@ -114,6 +115,12 @@ def train(config: ExperimentConfig):
                         dynamic_ncols=True)
         except ImportError:
             pbar = None
+    # one-batch-ahead pinned H2D prefetch on a copy stream (C5):
+    # the host-side loader + hipMemcpyAsync of step it+1 overlap step it's
+    # device compute; the compute stream only waits on the copy event.
+    prefetch = DevicePrefetcher(device)
+    pending = prefetch.start(*loader.batch("train", local_bs,
+                                           config.g_accum_iters))
     for it in range(first_step, config.max_steps):
         if it % config.eval_interval == 0:
             tl = evaluate(model, loader, "train", local_bs, device, eval_batches)
@@ -126,19 +133,25 @@ def train(config: ExperimentConfig):
                               warmup_steps=config.warmup_steps,
                               decay_steps=config.lr_decay_steps,
                               min_lr=config.min_lr)
-        x, y = loader.batch("train", local_bs, config.g_accum_iters)
+        x, y = prefetch.wait(pending)
 
         def one_step():
-            s = 0.0
+            nonlocal pending
+            losses = []
             for g in range(config.g_accum_iters):
-                xg = x[g].to(device, non_blocking=True)
-                yg = y[g].to(device, non_blocking=True)
-                loss = model.loss(xg, yg)
+                loss = model.loss(x[g], y[g])
                 loss.backward()
                 engine.microstep_end()
-                s += float(loss)
+                if g == 0 and it + 1 < config.max_steps:
+                    # next step's batch: host gather + async H2D, hidden
+                    # under the remaining microsteps' device work
+                    pending = prefetch.start(
+                        *loader.batch("train", local_bs, config.g_accum_iters))
+                losses.append(loss.detach())
             engine.step(lr, config.g_accum_iters)
-            return s / config.g_accum_iters
+            # single host sync per step (keeps the device queue deep
+            # through the microstep loop)
+            return float(torch.stack(losses).mean())
 
         if config.debug and it == first_step and config.rundir:
             # reference parity: --debug traces step 0 (src/train.py:205-211);
