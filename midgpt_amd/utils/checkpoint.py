@@ -137,10 +137,17 @@ def load_full_state(rundir: str, step: int | None = None):
         if not fn.startswith("rank"):
             continue
         sh = torch.load(os.path.join(d, fn), map_location="cpu", weights_only=False)
-        o, n = sh["shard_off"], sh["shard_size"]
-        master[o:o + n] = sh["master"]
-        m[o:o + n] = sh["m"]
-        v[o:o + n] = sh["v"]
+        if "pieces" in sh:
+            # bucketed shard: shard[po:po+n] lives at full[fo:fo+n]
+            for fo, n, po in sh["pieces"]:
+                master[fo:fo + n] = sh["master"][po:po + n]
+                m[fo:fo + n] = sh["m"][po:po + n]
+                v[fo:fo + n] = sh["v"][po:po + n]
+        else:
+            o, n = sh["shard_off"], sh["shard_size"]
+            master[o:o + n] = sh["master"]
+            m[o:o + n] = sh["m"]
+            v[o:o + n] = sh["v"]
         step_count = sh["step_count"]
     return {"step": step, "master": master, "m": m, "v": v,
             "step_count": step_count, "manifest": manifest}

@@ -52,9 +52,9 @@ def _worker(rank, world, init_file, zero, out_q, steps=3):
         model.loss(xs, ys).backward()
         engine.microstep_end()
         engine.step(1e-3)
-    # gather full master for comparison
+    # gather full master for comparison (piece map: bucketed shard layout)
     full = torch.zeros(engine.padded)
-    full[engine.shard_off:engine.shard_off + engine.shard_size] = engine.master
+    engine._scatter_shard(engine.master, full)
     if engine.zero:
         dist.all_reduce(full)
     # send as numpy bytes: torch tensors over mp queues use fd-passing,
@@ -83,6 +83,67 @@ def test_two_rank_matches_single_process(zero, tmp_path):
     for r, full in results.items():
         assert torch.allclose(full, ref[:full.numel()], atol=2e-5), \
             (r, (full - ref[:full.numel()]).abs().max())
+
+
+def _ckpt_worker(rank, world, init_file, rundir, out_q):
+    """ZeRO train -> sharded (bucketed-pieces) checkpoint -> rank 0
+    reassembles the full state and reloads it into a WORLD=1 engine."""
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["RANK"] = str(rank)
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    from midgpt_amd.models.gpt import GPT
+    from midgpt_amd.parallel.engine import ShardedAdamW
+    from midgpt_amd.utils import checkpoint as ckpt
+    torch.manual_seed(0)
+    model = GPT(TINY)
+    engine = ShardedAdamW(model, compute_dtype=torch.float32, zero=True)
+    x, y = _make_batch()
+    for _ in range(2):
+        model.loss(x[rank * 4:(rank + 1) * 4],
+                   y[rank * 4:(rank + 1) * 4]).backward()
+        engine.microstep_end()
+        engine.step(1e-3)
+    mngr = ckpt.CheckpointManager(rundir, save_interval=1)
+    mngr.save(2, engine)
+    mngr.wait()
+    full = torch.zeros(engine.padded)
+    engine._scatter_shard(engine.master, full)
+    dist.all_reduce(full)
+    ok = True
+    if rank == 0:
+        state = ckpt.load_full_state(rundir)
+        ok = torch.allclose(state["master"][:engine.total],
+                            full[:engine.total])
+        # reload into a fresh single-engine (resharding path)
+        m2 = GPT(TINY)
+        dist.destroy_process_group()  # world-1 engine below
+        del os.environ["WORLD_SIZE"]
+        e2 = ShardedAdamW(m2, compute_dtype=torch.float32, zero=True)
+        e2.load_state_full(state["master"], state["m"], state["v"],
+                           state["step_count"])
+        ok = ok and torch.allclose(e2.master[:engine.total],
+                                   full[:engine.total])
+    else:
+        dist.destroy_process_group()
+    out_q.put((rank, bool(ok)))
+
+
+def test_two_rank_zero_checkpoint_reshard(tmp_path):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    init_file = str(tmp_path / "pg_ck")
+    rundir = str(tmp_path / "run")
+    procs = [ctx.Process(target=_ckpt_worker, args=(r, 2, init_file, rundir, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for _ in range(2):
+        r, ok = q.get()
+        assert ok, f"rank {r} checkpoint reassembly mismatch"
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
 
 
 def _helper_worker(rank, world, init_file, out_q):
@@ -139,7 +200,7 @@ def _odd_worker(rank, world, init_file, out_q):
         engine.microstep_end()
         engine.step(1e-2)
     full = torch.zeros(engine.padded)
-    full[engine.shard_off:engine.shard_off + engine.shard_size] = engine.master
+    engine._scatter_shard(engine.master, full)
     dist.all_reduce(full)
     out_q.put((rank, full[:engine.total].numpy().copy()))
     dist.destroy_process_group()

@@ -157,8 +157,7 @@ def _two_proc_gpu_worker(rank, init_file, out_q):
         engine.step(1e-3)
     torch.cuda.synchronize()
     full = torch.zeros(engine.padded)
-    full[engine.shard_off:engine.shard_off + engine.shard_size] = \
-        engine.master.cpu()
+    engine._scatter_shard(engine.master.cpu(), full)
     dist.all_reduce(full)  # assemble the full master across shards (gloo)
     out_q.put((rank, full[:engine.total].numpy().copy()))
     dist.destroy_process_group()
