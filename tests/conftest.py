@@ -9,6 +9,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an MI355X GPU")
+    config.addinivalue_line(
+        "markers", "gpu8: requires >=2 GPUs (multi-GPU RCCL path)")
 
 
 def pytest_collection_modifyitems(config, items):
