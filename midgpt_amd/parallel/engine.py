@@ -161,11 +161,11 @@ class ShardedAdamW:
             for b in bl:
                 self._bucket_params[b] += 1
         self._hooks_on = self._overlap and self.n_buckets > 1
+        self._counts = list(self._bucket_params)
+        self._ready = [False] * self.n_buckets
+        self._next_launch = self.n_buckets - 1
+        self._launched = [False] * self.n_buckets
         if self._hooks_on:
-            self._counts = list(self._bucket_params)
-            self._ready = [False] * self.n_buckets
-            self._next_launch = self.n_buckets - 1
-            self._launched = [False] * self.n_buckets
             for i, p in enumerate(self.params):
                 p.register_post_accumulate_grad_hook(self._make_hook(i))
 
