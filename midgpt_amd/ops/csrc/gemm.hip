@@ -1,0 +1,271 @@
+// Hand-written CDNA4 bf16 GEMM (gfx950) — 256x256x64 tile, 8-wave,
+// glds-staged, 8-phase counted-vmcnt schedule (plan K6 upgrade: replace
+// hipBLASLt for the model's dominant shapes; reference counterpart:
+// XLA-generated matmuls from src/layers.py:55-57 / src/model.py:24-49).
+//
+// Geometry (guide cdna_hip_programming.md §5 "256² 8-phase template"):
+//   tile BM x BN = 256 x 256, K-step BK = 64, 8 waves (2M x 4N), 512 thr
+//   per-wave output 128 x 64 = acc f32x4[8][4] (128 VGPRs)
+//   MFMA: v_mfma_f32_16x16x32_bf16; 64 per wave per K-tile, in 4
+//   "quadrant" phases of 16
+//   LDS 128 KiB = 2 buffers x (A 32K + B 32K); each operand K-tile is
+//   staged as 2 halves of 128 rows x 64 k (16 KiB), 2 glds_dwordx4 per
+//   wave per half (8 waves cooperate)
+//   st_16x32 LDS swizzle: byte ^= ((byte>>9)&1)<<5 within each 1 KiB
+//   subtile == colbyte ^= 32 when (row & 4). glds writes lane-linear, so
+//   the swizzle is applied to the per-lane SOURCE address and to the
+//   ds_read address (both-sides involution, guide §5.4 rule 21).
+//
+// Schedule per K-tile u (2 raw s_barriers per phase, counted vmcnt only —
+// never __syncthreads, which would drain the in-flight glds):
+//   p0: ds_read B(u) [8xb128] + A strip0 [4]; glds A-lo(u+1)
+//   p1: ds_read A strip1;                     glds A-hi(u+1)
+//   p2: ds_read A strip2;                     glds B-lo(u+2)
+//   p3: ds_read A strip3;                     glds B-hi(u+2); vmcnt(4)
+//   The p3 vmcnt(4)+barrier is the guard for tile u+1's reads at the
+//   next p0 (A(u+1) = the 2 oldest of the 4 newest in-flight halves;
+//   B(u+1) landed earlier); the prologue ends with its own vmcnt(4) +
+//   barrier guarding tile 0.
+//   (each phase: ... ; s_barrier ; setprio(1) ; 16 MFMA ; setprio(0) ;
+//    s_barrier)
+// Slot-lifetime: A(u+1) overwrites A(u-1) (dead after (u-1).p3 barrier);
+// B(u+2) overwrites B(u) (dead after u.p0 barrier). vmcnt(4) at p3
+// guarantees A(u+1) landed before (u+1).p0's reads; vmcnt(6) at p0
+// guarantees A(u)+B(u) landed on the first tile (prologue stages 6
+// halves: A(0), B(0), B(1)).
+//
+// Operand layouts ("NT"): A[M,K] row-major, B[N,K] row-major, C[M,N]
+// row-major = A @ B^T — the natural layout of torch Linear fwd
+// (y = x @ W^T) AND of dgrad against a transposed weight copy
+// (dx = dy @ W = dy @ (W^T)^T). M, N, K must be multiples of 256/256/128.
+#include "common.h"
+
+using bf16x8g = __attribute__((ext_vector_type(8))) __bf16;
+
+DEVINL f32x4 mfma16(bf16x8g a, bf16x8g b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+#define GLDS(gsrc, lds_off) \
+  __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned*)(gsrc), \
+                                   (__attribute__((address_space(3))) unsigned*)(lds_off), 16, 0, 0)
+
+// LDS map (bytes): buf p (p=0,1) at p*65536; A at +0 (lo 0..16K, hi
+// 16..32K), B at +32768 (lo/hi).
+#define LDS_A(p) ((p) * 65536)
+#define LDS_B(p) ((p) * 65536 + 32768)
+
+// swizzled in-half byte offset for element (row, colbyte)
+DEVINL int swz(int row, int colbyte) {
+  return row * 128 + (colbyte ^ ((row & 4) << 3));
+}
+
+// ---------------------------------------------------------------------
+// Staging: half h (0/1) of operand tile u. 512 threads cover 16 KiB via
+// 16 wave-instructions (wave w: bytes [(2w+j)*1024, +1024)).
+// Per-lane global source honors the inverse swizzle.
+// rows0 = first global row of the half; ldb = row stride in BYTES.
+// ---------------------------------------------------------------------
+DEVINL void stage_half(const u16* __restrict__ g, long rows0, long ldb,
+                       long kbyte0, char* lds_base /*half base*/) {
+  const int w = wave_id();
+  const int l = lane_id();
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    int q = (w * 2 + j) * 1024 + l * 16;       // dest byte in half image
+    int row = q >> 7;
+    int colb = (q & 127) ^ ((row & 4) << 3);   // inverse swizzle on SOURCE
+    const char* src = (const char*)g + (rows0 + row) * ldb + kbyte0 + colb;
+    GLDS(src, lds_base + (w * 2 + j) * 1024);
+  }
+}
+
+// SAFE=1: drain-everything debug schedule (vmcnt(0) before every
+// barrier) — used to bisect schedule races from layout bugs.
+// ORDER: 0 = column-major tile walk (B-panel reuse), 1 = grouped walk
+// (GROUP tile_m rows per super-column: A panels L2-resident too).
+template <int SAFE, int GROUP>
+__launch_bounds__(512, 1)
+__global__ void gemm_nt_kernel(const u16* __restrict__ A,
+                               const u16* __restrict__ B,
+                               u16* __restrict__ C,
+                               int M, int N, int K, int swizzle_xcd) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int l = lane_id();
+  const int w = wave_id();
+  const int wave_m = w >> 2;       // 0..1
+  const int wave_n = w & 3;        // 0..3
+  const int tiles_m = M >> 8;
+  const int tiles_n = N >> 8;
+
+  int bid = blockIdx.x;
+  if (swizzle_xcd) {
+    // bijective XCD remap: contiguous chunk of the grid per XCD (T1)
+    const int nwg = tiles_m * tiles_n;
+    const int qq = nwg >> 3, rr = nwg & 7;
+    const int xcd = bid & 7, idx = bid >> 3;
+    bid = (xcd < rr ? xcd * (qq + 1) : rr * (qq + 1) + (xcd - rr) * qq) + idx;
+  }
+  // column-major walk: consecutive blocks share the B panel; GROUP>0
+  // walks n-fastest within GROUP-row bands so A panels stay L2-warm too
+  long bm, bn;
+  if (GROUP > 0) {
+    const int band = GROUP * tiles_n;      // blocks per band
+    const int b0 = bid / band, r0 = bid % band;
+    bm = (long)(b0 * GROUP + r0 % GROUP) << 8;
+    bn = (long)(r0 / GROUP) << 8;
+  } else {
+    bm = (long)(bid % tiles_m) << 8;
+    bn = (long)(bid / tiles_m) << 8;
+  }
+
+  const long ldab = (long)K * 2;   // A/B row stride bytes
+  const int NT = K >> 6;           // K-tiles
+
+  f32x4 acc[8][4] = {};
+
+  // ---- prologue: stage A(0), B(0), B(1) --------------------------------
+  stage_half(A, bm + 0, ldab, 0, smem + LDS_A(0) + 0);
+  stage_half(A, bm + 128, ldab, 0, smem + LDS_A(0) + 16384);
+  stage_half(B, bn + 0, ldab, 0, smem + LDS_B(0) + 0);
+  stage_half(B, bn + 128, ldab, 0, smem + LDS_B(0) + 16384);
+  if (NT > 1) {
+    stage_half(B, bn + 0, ldab, 128, smem + LDS_B(1) + 0);
+    stage_half(B, bn + 128, ldab, 128, smem + LDS_B(1) + 16384);
+  }
+  // A(0)+B(0) (the first 8 glds) must land before tile 0's ds_reads;
+  // B(1) (the last 4) may stay in flight. Counted wait + raw barrier —
+  // the steady-state guard for tile u+1 is phase 3's vmcnt(4)+barrier.
+  if (SAFE || NT == 1) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  // per-lane read offsets
+  const int frag_colb = (l >> 4) * 16;        // k-chunk byte within kstep
+  const int a_row = l & 15;                   // row within 16-row fragment
+  // B fragments: n_local = wave_n*64 + nr*16 + (l&15)
+  const int b_nloc = wave_n * 64 + (l & 15);
+  char* const lds = smem;
+
+  bf16x8g bfr[4][2];   // B frags [nr][kstep], live across the tile
+  bf16x8g afr[2][2];   // A strip frags [mr][kstep], per quadrant phase
+
+#define READ_A(q, ks, mr)                                                     \
+  afr[mr][ks] = *(const bf16x8g*)(lds + LDS_A(par) + wave_m * 16384 +         \
+      swz((q) * 32 + (mr) * 16 + a_row, (ks) * 64 + frag_colb))
+#define READ_B(nr, ks)                                                        \
+  bfr[nr][ks] = *(const bf16x8g*)(lds + LDS_B(par) +                          \
+      ((b_nloc + (nr) * 16) >> 7) * 16384 +                                   \
+      swz((b_nloc + (nr) * 16) & 127, (ks) * 64 + frag_colb))
+#define MFMA_QUAD(q)                                                          \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                            \
+    _Pragma("unroll") for (int mr = 0; mr < 2; ++mr)                          \
+      _Pragma("unroll") for (int nr = 0; nr < 4; ++nr)                        \
+        acc[(q) * 2 + mr][nr] = mfma16(afr[mr][ks], bfr[nr][ks],              \
+                                       acc[(q) * 2 + mr][nr])
+// Raw s_barrier is NOT a compiler memory fence: without the empty
+// "memory"-clobber asm on both sides hipcc may hoist a glds / ds_read
+// across it, re-staging an LDS slot other waves still read (observed as
+// a rare nondeterministic ~1e-3-per-block corruption). The fences order
+// the compiler only; the hardware wait discipline stays counted-vmcnt.
+#define BAR() do { if (SAFE) asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); \
+                  asm volatile("" ::: "memory");                              \
+                  __builtin_amdgcn_s_barrier();                               \
+                  asm volatile("" ::: "memory"); } while (0)
+#define PRIO(x) __builtin_amdgcn_s_setprio(x)
+#define VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
+
+  for (int u = 0; u < NT; ++u) {
+    const int par = u & 1;
+    const int nxt = par ^ 1;
+    const long kb1 = (long)(u + 1) << 7;   // byte col of K-tile u+1
+    const long kb2 = (long)(u + 2) << 7;
+
+    // ---- phase 0: B(u) full + A strip 0; stage A-lo(u+1) ----
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) { READ_B(nr, 0); READ_B(nr, 1); }
+    READ_A(0, 0, 0); READ_A(0, 1, 0); READ_A(0, 0, 1); READ_A(0, 1, 1);
+    if (u + 1 < NT) stage_half(A, bm + 0, ldab, kb1, lds + LDS_A(nxt) + 0);
+    BAR();
+    PRIO(1); MFMA_QUAD(0); PRIO(0);
+    BAR();
+    // ---- phase 1: A strip 1; stage A-hi(u+1) ----
+    READ_A(1, 0, 0); READ_A(1, 1, 0); READ_A(1, 0, 1); READ_A(1, 1, 1);
+    if (u + 1 < NT) stage_half(A, bm + 128, ldab, kb1, lds + LDS_A(nxt) + 16384);
+    BAR();
+    PRIO(1); MFMA_QUAD(1); PRIO(0);
+    BAR();
+    // ---- phase 2: A strip 2; stage B-lo(u+2) ----
+    READ_A(2, 0, 0); READ_A(2, 1, 0); READ_A(2, 0, 1); READ_A(2, 1, 1);
+    if (u + 2 < NT) stage_half(B, bn + 0, ldab, kb2, lds + LDS_B(par) + 0);
+    BAR();
+    PRIO(1); MFMA_QUAD(2); PRIO(0);
+    BAR();
+    // ---- phase 3: A strip 3; stage B-hi(u+2) ----
+    READ_A(3, 0, 0); READ_A(3, 1, 0); READ_A(3, 0, 1); READ_A(3, 1, 1);
+    if (u + 2 < NT) {
+      stage_half(B, bn + 128, ldab, kb2, lds + LDS_B(par) + 16384);
+      VMCNT(4);  // leaves exactly B(u+2)'s 4 glds; A(u+1) landed
+    } else {
+      // tail: B(u+2) skipped, so the 4 newest in-flight glds would be
+      // A(u+1) itself — drain fully before tile u+1 reads it
+      VMCNT(0);
+    }
+    BAR();
+    PRIO(1); MFMA_QUAD(3); PRIO(0);
+    BAR();
+  }
+#undef READ_A
+#undef READ_B
+#undef MFMA_QUAD
+
+  // ---- epilogue: bf16 C write (D layout: col=l&15, row=4*(l>>4)+r) ----
+  const long ldc = N;
+  const long crow0 = bm + wave_m * 128 + (l >> 4) * 4;
+  const long ccol0 = bn + wave_n * 64 + (l & 15);
+#pragma unroll
+  for (int mr = 0; mr < 8; ++mr) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long row = crow0 + mr * 16 + r;
+#pragma unroll
+      for (int nr = 0; nr < 4; ++nr)
+        C[row * ldc + ccol0 + nr * 16] = f2b(acc[mr][nr][r]);
+    }
+  }
+}
+
+// host-side launcher (shared by bindings and the standalone probe)
+template <int SAFE, int GROUP>
+static hipError_t launch_gemm_nt_t(const u16* A, const u16* B, u16* C,
+                                   int M, int N, int K, hipStream_t stream,
+                                   int swizzle_xcd) {
+  if (M % 256 || N % 256 || K % 64) return hipErrorInvalidValue;
+  if (GROUP > 0 && (M >> 8) % GROUP) return hipErrorInvalidValue;
+  static int lds_set = 0;
+  if (!lds_set) {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gemm_nt_kernel<SAFE, GROUP>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+    lds_set = 1;
+  }
+  const int grid = (M >> 8) * (N >> 8);
+  hipLaunchKernelGGL((gemm_nt_kernel<SAFE, GROUP>), dim3(grid), dim3(512),
+                     131072, stream, A, B, C, M, N, K, swizzle_xcd);
+  return hipGetLastError();
+}
+
+static inline hipError_t launch_gemm_nt(const u16* A, const u16* B, u16* C,
+                                        int M, int N, int K,
+                                        hipStream_t stream,
+                                        int swizzle_xcd = 1, int safe = 0,
+                                        int group = 0) {
+  if (safe) {
+    if (group == 8) return launch_gemm_nt_t<1, 8>(A, B, C, M, N, K, stream, swizzle_xcd);
+    return launch_gemm_nt_t<1, 0>(A, B, C, M, N, K, stream, swizzle_xcd);
+  }
+  if (group == 8) return launch_gemm_nt_t<0, 8>(A, B, C, M, N, K, stream, swizzle_xcd);
+  if (group == 16) return launch_gemm_nt_t<0, 16>(A, B, C, M, N, K, stream, swizzle_xcd);
+  return launch_gemm_nt_t<0, 0>(A, B, C, M, N, K, stream, swizzle_xcd);
+}

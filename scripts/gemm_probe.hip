@@ -1,0 +1,116 @@
+// Standalone probe for the 256x256x64 8-phase bf16 GEMM (gemm.hip):
+// numerics refcheck (host fp32 on sampled rows) + TF/s timing.
+// Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 -I. \
+//          scripts/gemm_probe.hip -o gpurun_out/gemm_probe
+// Run:   ./gemm_probe M N K [iters]
+#include <cstdio>
+#include <cstdlib>
+#include <cmath>
+#include <vector>
+#include <random>
+
+#define TORCH_CHECK(cond, ...) do { if (!(cond)) { fprintf(stderr, "check failed\n"); abort(); } } while (0)
+#include "midgpt_amd/ops/csrc/gemm.hip"
+
+#define HC(x) do { hipError_t e = (x); if (e != hipSuccess) { \
+  fprintf(stderr, "HIP error %s at %s:%d\n", hipGetErrorString(e), __FILE__, __LINE__); exit(1);} } while (0)
+
+static inline u16 h_f2b(float f) {
+  union { float f; unsigned u; } c; c.f = f;
+  unsigned u = c.u;
+  unsigned r = u + 0x7fffu + ((u >> 16) & 1u);
+  return (u16)(r >> 16);
+}
+static inline float h_b2f(u16 h) {
+  union { unsigned u; float f; } c; c.u = ((unsigned)h) << 16; return c.f;
+}
+
+int main(int argc, char** argv) {
+  int M = argc > 1 ? atoi(argv[1]) : 4096;
+  int N = argc > 2 ? atoi(argv[2]) : 4096;
+  int K = argc > 3 ? atoi(argv[3]) : 4096;
+  int iters = argc > 4 ? atoi(argv[4]) : 20;
+  int swz_xcd = argc > 5 ? atoi(argv[5]) : 1;
+  int safe = argc > 6 ? atoi(argv[6]) : 0;
+  int group = argc > 7 ? atoi(argv[7]) : 0;
+
+  std::mt19937 rng(1234);
+  std::uniform_real_distribution<float> dist(-1.f, 1.f);
+  std::vector<u16> hA((size_t)M * K), hB((size_t)N * K);
+  for (auto& x : hA) x = h_f2b(dist(rng));
+  for (auto& x : hB) x = h_f2b(dist(rng));
+
+  u16 *dA, *dB, *dC;
+  HC(hipMalloc(&dA, hA.size() * 2));
+  HC(hipMalloc(&dB, hB.size() * 2));
+  HC(hipMalloc(&dC, (size_t)M * N * 2));
+  HC(hipMemcpy(dA, hA.data(), hA.size() * 2, hipMemcpyHostToDevice));
+  HC(hipMemcpy(dB, hB.data(), hB.size() * 2, hipMemcpyHostToDevice));
+
+  HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group));
+  HC(hipDeviceSynchronize());
+
+  // refcheck: 16 sampled rows, full N, host fp32 accumulate
+  std::vector<u16> hC((size_t)M * N);
+  HC(hipMemcpy(hC.data(), dC, hC.size() * 2, hipMemcpyDeviceToHost));
+  double max_rel = 0, max_abs = 0;
+  int bad = 0;
+  std::uniform_int_distribution<int> rowd(0, M - 1);
+  for (int s = 0; s < 16; ++s) {
+    int m = rowd(rng);
+    for (int n = 0; n < N; ++n) {
+      float ref = 0;
+      for (int k = 0; k < K; ++k)
+        ref += h_b2f(hA[(size_t)m * K + k]) * h_b2f(hB[(size_t)n * K + k]);
+      float got = h_b2f(hC[(size_t)m * N + n]);
+      double a = fabs(got - ref);
+      double r = a / (fabs(ref) + 1e-3);
+      if (r > max_rel) max_rel = r;
+      if (a > max_abs) max_abs = a;
+      // bf16 out + fp32 accum vs host fp32: loose elementwise bound
+      if (a > 0.05 * sqrt((double)K) && r > 0.05) {
+        if (bad < 8)
+          printf("  bad m=%d n=%d (tile %d,%d in-tile %d,%d) got %.4f ref %.4f\n",
+                 m, n, m >> 8, n >> 8, m & 255, n & 255, got, ref);
+        ++bad;
+      }
+    }
+  }
+  printf("refcheck: max_rel %.3e max_abs %.3e bad %d (16 rows x all cols)\n",
+         max_rel, max_abs, bad);
+  if (bad > 0) { printf("FAIL\n"); return 1; }
+
+  // determinism: re-run DET times, byte-compare against the first C
+  int det = argc > 8 ? atoi(argv[8]) : 3;
+  std::vector<u16> hC2((size_t)M * N);
+  int ndet = 0;
+  for (int d = 0; d < det; ++d) {
+    HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group));
+    HC(hipDeviceSynchronize());
+    HC(hipMemcpy(hC2.data(), dC, hC2.size() * 2, hipMemcpyDeviceToHost));
+    size_t diff = 0;
+    for (size_t i = 0; i < hC2.size(); ++i) diff += (hC2[i] != hC[i]);
+    if (diff) { ++ndet; printf("  nondet run %d: %zu elems differ\n", d, diff); }
+  }
+  if (ndet) { printf("NONDETERMINISTIC (%d/%d runs)\nFAIL\n", ndet, det); return 1; }
+
+  // timing
+  hipEvent_t t0, t1;
+  HC(hipEventCreate(&t0));
+  HC(hipEventCreate(&t1));
+  for (int i = 0; i < 3; ++i)
+    HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group));
+  HC(hipDeviceSynchronize());
+  HC(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i)
+    HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group));
+  HC(hipEventRecord(t1));
+  HC(hipEventSynchronize(t1));
+  float ms;
+  HC(hipEventElapsedTime(&ms, t0, t1));
+  double tf = 2.0 * M * N * K * iters / (ms * 1e-3) / 1e12;
+  printf("gemm_nt %dx%dx%d swz%d safe%d grp%d: %.3f ms/iter  %.1f TF/s\n",
+         M, N, K, swz_xcd, safe, group, ms / iters, tf);
+  printf("PASS\n");
+  return 0;
+}
