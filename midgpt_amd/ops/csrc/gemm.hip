@@ -55,9 +55,13 @@ DEVINL f32x4 mfma16(bf16x8g a, bf16x8g b, f32x4 c) {
 #define LDS_A(p) ((p) * 65536)
 #define LDS_B(p) ((p) * 65536 + 32768)
 
-// swizzled in-half byte offset for element (row, colbyte)
+// swizzled in-half byte offset for element (row, colbyte).
+// T2 XOR swizzle (row&7)<<4: a 16-lane ds_read_b128 group reading 16
+// consecutive rows at one column lands on 8 distinct bank slots (2-way)
+// instead of 4-way with a single toggled bit; involution within each
+// 128-byte row so the glds source-side permutation stays 16B-contiguous.
 DEVINL int swz(int row, int colbyte) {
-  return row * 128 + (colbyte ^ ((row & 4) << 3));
+  return row * 128 + (colbyte ^ ((row & 7) << 4));
 }
 
 // ---------------------------------------------------------------------
@@ -74,7 +78,7 @@ DEVINL void stage_half(const u16* __restrict__ g, long rows0, long ldb,
   for (int j = 0; j < 2; ++j) {
     int q = (w * 2 + j) * 1024 + l * 16;       // dest byte in half image
     int row = q >> 7;
-    int colb = (q & 127) ^ ((row & 4) << 3);   // inverse swizzle on SOURCE
+    int colb = (q & 127) ^ ((row & 7) << 4);   // inverse swizzle on SOURCE
     const char* src = (const char*)g + (rows0 + row) * ldb + kbyte0 + colb;
     GLDS(src, lds_base + (w * 2 + j) * 1024);
   }
@@ -84,7 +88,7 @@ DEVINL void stage_half(const u16* __restrict__ g, long rows0, long ldb,
 // barrier) — used to bisect schedule races from layout bugs.
 // ORDER: 0 = column-major tile walk (B-panel reuse), 1 = grouped walk
 // (GROUP tile_m rows per super-column: A panels L2-resident too).
-template <int SAFE, int GROUP>
+template <int SAFE, int GROUP, int PH>
 __launch_bounds__(512, 1)
 __global__ void gemm_nt_kernel(const u16* __restrict__ A,
                                const u16* __restrict__ B,
@@ -149,20 +153,23 @@ __global__ void gemm_nt_kernel(const u16* __restrict__ A,
   char* const lds = smem;
 
   bf16x8g bfr[4][2];   // B frags [nr][kstep], live across the tile
-  bf16x8g afr[2][2];   // A strip frags [mr][kstep], per quadrant phase
+  bf16x8g afr[4][2];   // A strip frags: two [mr][ks] pairs (double-buffer)
 
-#define READ_A(q, ks, mr)                                                     \
-  afr[mr][ks] = *(const bf16x8g*)(lds + LDS_A(par) + wave_m * 16384 +         \
-      swz((q) * 32 + (mr) * 16 + a_row, (ks) * 64 + frag_colb))
-#define READ_B(nr, ks)                                                        \
-  bfr[nr][ks] = *(const bf16x8g*)(lds + LDS_B(par) +                          \
+// slot: afr register slot; the fragment is rows (q*32 + (slot&1)*16)
+#define READ_AP(pp, q, ks, slot)                                              \
+  afr[slot][ks] = *(const bf16x8g*)(lds + LDS_A(pp) + wave_m * 16384 +        \
+      swz((q) * 32 + ((slot) & 1) * 16 + a_row, (ks) * 64 + frag_colb))
+#define READ_A(q, ks, slot) READ_AP(par, q, ks, slot)
+#define READ_BP(pp, nr, ks)                                                   \
+  bfr[nr][ks] = *(const bf16x8g*)(lds + LDS_B(pp) +                           \
       ((b_nloc + (nr) * 16) >> 7) * 16384 +                                   \
       swz((b_nloc + (nr) * 16) & 127, (ks) * 64 + frag_colb))
-#define MFMA_QUAD(q)                                                          \
+#define READ_B(nr, ks) READ_BP(par, nr, ks)
+#define MFMA_QUAD2(q, base)                                                   \
   _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                            \
     _Pragma("unroll") for (int mr = 0; mr < 2; ++mr)                          \
       _Pragma("unroll") for (int nr = 0; nr < 4; ++nr)                        \
-        acc[(q) * 2 + mr][nr] = mfma16(afr[mr][ks], bfr[nr][ks],              \
+        acc[(q) * 2 + mr][nr] = mfma16(afr[(base) + mr][ks], bfr[nr][ks],     \
                                        acc[(q) * 2 + mr][nr])
 // Raw s_barrier is NOT a compiler memory fence: without the empty
 // "memory"-clobber asm on both sides hipcc may hoist a glds / ds_read
@@ -176,44 +183,109 @@ __global__ void gemm_nt_kernel(const u16* __restrict__ A,
 #define PRIO(x) __builtin_amdgcn_s_setprio(x)
 #define VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
 
+  if constexpr (PH == 4) {
+    // tile 0's fragments (the loop reads tile u+1's at u.p3)
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) { READ_BP(0, nr, 0); READ_BP(0, nr, 1); }
+    READ_AP(0, 0, 0, 0); READ_AP(0, 0, 1, 0);
+    READ_AP(0, 0, 0, 1); READ_AP(0, 0, 1, 1);
+  }
+
   for (int u = 0; u < NT; ++u) {
     const int par = u & 1;
     const int nxt = par ^ 1;
     const long kb1 = (long)(u + 1) << 7;   // byte col of K-tile u+1
     const long kb2 = (long)(u + 2) << 7;
 
-    // ---- phase 0: B(u) full + A strip 0; stage A-lo(u+1) ----
+    if constexpr (PH == 2) {
+      // ---- 2 phases per tile: longer MFMA clusters, half the barriers.
+      // Same vmcnt discipline (stage order and per-wave glds counts are
+      // identical to the 4-phase schedule, just paired).
+      // phase A: B(u) + A strips 0,1; stage A(u+1) (both halves)
 #pragma unroll
-    for (int nr = 0; nr < 4; ++nr) { READ_B(nr, 0); READ_B(nr, 1); }
-    READ_A(0, 0, 0); READ_A(0, 1, 0); READ_A(0, 0, 1); READ_A(0, 1, 1);
+      for (int nr = 0; nr < 4; ++nr) { READ_B(nr, 0); READ_B(nr, 1); }
+      READ_A(0, 0, 0); READ_A(0, 1, 0); READ_A(0, 0, 1); READ_A(0, 1, 1);
+      READ_A(1, 0, 2); READ_A(1, 1, 2); READ_A(1, 0, 3); READ_A(1, 1, 3);
+      if (u + 1 < NT) {
+        stage_half(A, bm + 0, ldab, kb1, lds + LDS_A(nxt) + 0);
+        stage_half(A, bm + 128, ldab, kb1, lds + LDS_A(nxt) + 16384);
+      }
+      BAR();
+      PRIO(1);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+          for (int nr = 0; nr < 4; ++nr)
+            acc[mr][nr] = mfma16(afr[mr][ks], bfr[nr][ks], acc[mr][nr]);
+      PRIO(0);
+      BAR();
+      // phase B: A strips 2,3; stage B(u+2)
+      READ_A(2, 0, 0); READ_A(2, 1, 0); READ_A(2, 0, 1); READ_A(2, 1, 1);
+      READ_A(3, 0, 2); READ_A(3, 1, 2); READ_A(3, 0, 3); READ_A(3, 1, 3);
+      if (u + 2 < NT) {
+        stage_half(B, bn + 0, ldab, kb2, lds + LDS_B(par) + 0);
+        stage_half(B, bn + 128, ldab, kb2, lds + LDS_B(par) + 16384);
+        VMCNT(4);
+      } else {
+        VMCNT(0);
+      }
+      BAR();
+      PRIO(1);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+          for (int nr = 0; nr < 4; ++nr)
+            acc[4 + mr][nr] = mfma16(afr[mr][ks], bfr[nr][ks], acc[4 + mr][nr]);
+      PRIO(0);
+      BAR();
+      continue;
+    }
+
+    // ---- 4 phases per tile, POST-MFMA fragment prefetch: phase q's
+    // MFMA consumes afr pair (q&1); the reads for phase q+1 are issued
+    // right after the MFMA cluster into the other pair, so their LDS
+    // latency hides under the barrier + next phase's stage segment.
+    // B(u+1)+strip0(u+1) reads go after p3's vmcnt(4)+MFMA (guarded).
+    // phase 0: stage A-lo(u+1); MFMA q0 [afr pair0]; read strip1->pair1
     if (u + 1 < NT) stage_half(A, bm + 0, ldab, kb1, lds + LDS_A(nxt) + 0);
     BAR();
-    PRIO(1); MFMA_QUAD(0); PRIO(0);
+    PRIO(1); MFMA_QUAD2(0, 0); PRIO(0);
+    READ_A(1, 0, 2); READ_A(1, 1, 2); READ_A(1, 0, 3); READ_A(1, 1, 3);
     BAR();
-    // ---- phase 1: A strip 1; stage A-hi(u+1) ----
-    READ_A(1, 0, 0); READ_A(1, 1, 0); READ_A(1, 0, 1); READ_A(1, 1, 1);
+    // phase 1: stage A-hi(u+1); MFMA q1 [pair1]; read strip2->pair0
     if (u + 1 < NT) stage_half(A, bm + 128, ldab, kb1, lds + LDS_A(nxt) + 16384);
     BAR();
-    PRIO(1); MFMA_QUAD(1); PRIO(0);
-    BAR();
-    // ---- phase 2: A strip 2; stage B-lo(u+2) ----
+    PRIO(1); MFMA_QUAD2(1, 2); PRIO(0);
     READ_A(2, 0, 0); READ_A(2, 1, 0); READ_A(2, 0, 1); READ_A(2, 1, 1);
+    BAR();
+    // phase 2: stage B-lo(u+2); MFMA q2 [pair0]; read strip3->pair1
     if (u + 2 < NT) stage_half(B, bn + 0, ldab, kb2, lds + LDS_B(par) + 0);
     BAR();
-    PRIO(1); MFMA_QUAD(2); PRIO(0);
+    PRIO(1); MFMA_QUAD2(2, 0); PRIO(0);
+    READ_A(3, 0, 2); READ_A(3, 1, 2); READ_A(3, 0, 3); READ_A(3, 1, 3);
     BAR();
-    // ---- phase 3: A strip 3; stage B-hi(u+2) ----
-    READ_A(3, 0, 0); READ_A(3, 1, 0); READ_A(3, 0, 1); READ_A(3, 1, 1);
+    // phase 3: stage B-hi(u+2); vmcnt; MFMA q3 [pair1]; read B(u+1) +
+    // strip0(u+1) -> pair0 (safe: A(u+1)/B(u+1) landed per the vmcnt)
     if (u + 2 < NT) {
       stage_half(B, bn + 128, ldab, kb2, lds + LDS_B(par) + 16384);
-      VMCNT(4);  // leaves exactly B(u+2)'s 4 glds; A(u+1) landed
+      VMCNT(4);  // leaves exactly B(u+2)'s 4 glds; A(u+1)/B(u+1) landed
     } else {
       // tail: B(u+2) skipped, so the 4 newest in-flight glds would be
       // A(u+1) itself — drain fully before tile u+1 reads it
       VMCNT(0);
     }
     BAR();
-    PRIO(1); MFMA_QUAD(3); PRIO(0);
+    PRIO(1); MFMA_QUAD2(3, 2); PRIO(0);
+    if (u + 1 < NT) {
+#pragma unroll
+      for (int nr = 0; nr < 4; ++nr) { READ_BP(nxt, nr, 0); READ_BP(nxt, nr, 1); }
+      READ_AP(nxt, 0, 0, 0); READ_AP(nxt, 0, 1, 0);
+      READ_AP(nxt, 0, 0, 1); READ_AP(nxt, 0, 1, 1);
+    }
     BAR();
   }
 #undef READ_A
@@ -236,8 +308,165 @@ __global__ void gemm_nt_kernel(const u16* __restrict__ A,
   }
 }
 
-// host-side launcher (shared by bindings and the standalone probe)
+
+// ---------------------------------------------------------------------
+// 32x32x16-MFMA sibling: same tile/LDS/staging/schedule, half the MFMA
+// instructions (32768 FLOP per issue slot vs 16384) — tests whether the
+// 16x16 variant is issue-bound. Wave tile 128x64 = 4 m_reps x 2 n_reps
+// of 32x32 fragments; K-tile = 4 ksteps of 16.
+// ---------------------------------------------------------------------
+using bf16x8m = bf16x8g;
+DEVINL f32x16 mfma32(bf16x8m a, bf16x8m b, f32x16 c) {
+  return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
 template <int SAFE, int GROUP>
+__launch_bounds__(512, 1)
+__global__ void gemm_nt32_kernel(const u16* __restrict__ A,
+                                 const u16* __restrict__ B,
+                                 u16* __restrict__ C,
+                                 int M, int N, int K, int swizzle_xcd) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int l = lane_id();
+  const int w = wave_id();
+  const int wave_m = w >> 2;
+  const int wave_n = w & 3;
+  const int tiles_m = M >> 8;
+  const int tiles_n = N >> 8;
+
+  int bid = blockIdx.x;
+  if (swizzle_xcd) {
+    const int nwg = tiles_m * tiles_n;
+    const int qq = nwg >> 3, rr = nwg & 7;
+    const int xcd = bid & 7, idx = bid >> 3;
+    bid = (xcd < rr ? xcd * (qq + 1) : rr * (qq + 1) + (xcd - rr) * qq) + idx;
+  }
+  long bm, bn;
+  if (GROUP > 0) {
+    const int band = GROUP * tiles_n;
+    const int b0 = bid / band, r0 = bid % band;
+    bm = (long)(b0 * GROUP + r0 % GROUP) << 8;
+    bn = (long)(r0 / GROUP) << 8;
+  } else {
+    bm = (long)(bid % tiles_m) << 8;
+    bn = (long)(bid / tiles_m) << 8;
+  }
+
+  const long ldab = (long)K * 2;
+  const int NT = K >> 6;
+
+  f32x16 acc[4][2] = {};
+
+  stage_half(A, bm + 0, ldab, 0, smem + LDS_A(0) + 0);
+  stage_half(A, bm + 128, ldab, 0, smem + LDS_A(0) + 16384);
+  stage_half(B, bn + 0, ldab, 0, smem + LDS_B(0) + 0);
+  stage_half(B, bn + 128, ldab, 0, smem + LDS_B(0) + 16384);
+  if (NT > 1) {
+    stage_half(B, bn + 0, ldab, 128, smem + LDS_B(1) + 0);
+    stage_half(B, bn + 128, ldab, 128, smem + LDS_B(1) + 16384);
+  }
+  if (SAFE || NT == 1) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  // 32x32 fragment addressing: A row l&31, k-chunk byte 16*(l>>5)
+  const int a_row = l & 31;
+  const int kchunk = (l >> 5) * 16;
+  const int b_nloc = wave_n * 64 + (l & 31);
+  char* const lds = smem;
+
+  bf16x8m bfr[2][4];    // [nr][kstep]
+  bf16x8m afr[2][4];    // two buffers x 4 ksteps (one m_rep each)
+
+#define RD_A32(pp, q, ks, buf)                                                \
+  afr[buf][ks] = *(const bf16x8m*)(lds + LDS_A(pp) + wave_m * 16384 +         \
+      swz((q) * 32 + a_row, (ks) * 32 + kchunk))
+#define RD_B32(pp, nr, ks)                                                    \
+  bfr[nr][ks] = *(const bf16x8m*)(lds + LDS_B(pp) +                           \
+      ((b_nloc + (nr) * 32) >> 7) * 16384 +                                   \
+      swz((b_nloc + (nr) * 32) & 127, (ks) * 32 + kchunk))
+#define MFMA_M32(q, buf)                                                      \
+  _Pragma("unroll") for (int ks = 0; ks < 4; ++ks)                            \
+    _Pragma("unroll") for (int nr = 0; nr < 2; ++nr)                          \
+      acc[q][nr] = mfma32(afr[buf][ks], bfr[nr][ks], acc[q][nr])
+#define BAR32() do { if (SAFE) asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); \
+                  asm volatile("" ::: "memory");                              \
+                  __builtin_amdgcn_s_barrier();                               \
+                  asm volatile("" ::: "memory"); } while (0)
+
+  // tile 0 fragments
+#pragma unroll
+  for (int nr = 0; nr < 2; ++nr)
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) RD_B32(0, nr, ks);
+  RD_A32(0, 0, 0, 0); RD_A32(0, 0, 1, 0); RD_A32(0, 0, 2, 0); RD_A32(0, 0, 3, 0);
+
+  for (int u = 0; u < NT; ++u) {
+    const int par = u & 1;
+    const int nxt = par ^ 1;
+    const long kb1 = (long)(u + 1) << 7;
+    const long kb2 = (long)(u + 2) << 7;
+
+    // p0: m_rep 0
+    if (u + 1 < NT) stage_half(A, bm + 0, ldab, kb1, lds + LDS_A(nxt) + 0);
+    BAR32();
+    PRIO(1); MFMA_M32(0, 0); PRIO(0);
+    RD_A32(par, 1, 0, 1); RD_A32(par, 1, 1, 1); RD_A32(par, 1, 2, 1); RD_A32(par, 1, 3, 1);
+    BAR32();
+    // p1: m_rep 1
+    if (u + 1 < NT) stage_half(A, bm + 128, ldab, kb1, lds + LDS_A(nxt) + 16384);
+    BAR32();
+    PRIO(1); MFMA_M32(1, 1); PRIO(0);
+    RD_A32(par, 2, 0, 0); RD_A32(par, 2, 1, 0); RD_A32(par, 2, 2, 0); RD_A32(par, 2, 3, 0);
+    BAR32();
+    // p2: m_rep 2
+    if (u + 2 < NT) stage_half(B, bn + 0, ldab, kb2, lds + LDS_B(par) + 0);
+    BAR32();
+    PRIO(1); MFMA_M32(2, 0); PRIO(0);
+    RD_A32(par, 3, 0, 1); RD_A32(par, 3, 1, 1); RD_A32(par, 3, 2, 1); RD_A32(par, 3, 3, 1);
+    BAR32();
+    // p3: m_rep 3; then prefetch tile u+1 fragments
+    if (u + 2 < NT) {
+      stage_half(B, bn + 128, ldab, kb2, lds + LDS_B(par) + 16384);
+      VMCNT(4);
+    } else {
+      VMCNT(0);
+    }
+    BAR32();
+    PRIO(1); MFMA_M32(3, 1); PRIO(0);
+    if (u + 1 < NT) {
+#pragma unroll
+      for (int nr = 0; nr < 2; ++nr)
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) RD_B32(nxt, nr, ks);
+      RD_A32(nxt, 0, 0, 0); RD_A32(nxt, 0, 1, 0); RD_A32(nxt, 0, 2, 0); RD_A32(nxt, 0, 3, 0);
+    }
+    BAR32();
+  }
+#undef RD_A32
+#undef RD_B32
+#undef MFMA_M32
+#undef BAR32
+
+  // epilogue: D map col = l&31, row = (r&3) + 8*(r>>2) + 4*(l>>5)
+  const long ldc = N;
+  const long ccol0 = bn + wave_n * 64 + (l & 31);
+#pragma unroll
+  for (int mrep = 0; mrep < 4; ++mrep) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const long row = bm + wave_m * 128 + mrep * 32 +
+                       (r & 3) + 8 * (r >> 2) + 4 * (l >> 5);
+#pragma unroll
+      for (int nr = 0; nr < 2; ++nr)
+        C[row * ldc + ccol0 + nr * 32] = f2b(acc[mrep][nr][r]);
+    }
+  }
+}
+
+// host-side launcher (shared by bindings and the standalone probe)
+template <int SAFE, int GROUP, int PH>
 static hipError_t launch_gemm_nt_t(const u16* A, const u16* B, u16* C,
                                    int M, int N, int K, hipStream_t stream,
                                    int swizzle_xcd) {
@@ -246,12 +475,31 @@ static hipError_t launch_gemm_nt_t(const u16* A, const u16* B, u16* C,
   static int lds_set = 0;
   if (!lds_set) {
     (void)hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&gemm_nt_kernel<SAFE, GROUP>),
+        reinterpret_cast<const void*>(&gemm_nt_kernel<SAFE, GROUP, PH>),
         hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
     lds_set = 1;
   }
   const int grid = (M >> 8) * (N >> 8);
-  hipLaunchKernelGGL((gemm_nt_kernel<SAFE, GROUP>), dim3(grid), dim3(512),
+  hipLaunchKernelGGL((gemm_nt_kernel<SAFE, GROUP, PH>), dim3(grid), dim3(512),
+                     131072, stream, A, B, C, M, N, K, swizzle_xcd);
+  return hipGetLastError();
+}
+
+template <int SAFE, int GROUP>
+static hipError_t launch_gemm_nt32_t(const u16* A, const u16* B, u16* C,
+                                     int M, int N, int K, hipStream_t stream,
+                                     int swizzle_xcd) {
+  if (M % 256 || N % 256 || K % 64) return hipErrorInvalidValue;
+  if (GROUP > 0 && (M >> 8) % GROUP) return hipErrorInvalidValue;
+  static int lds_set = 0;
+  if (!lds_set) {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gemm_nt32_kernel<SAFE, GROUP>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+    lds_set = 1;
+  }
+  const int grid = (M >> 8) * (N >> 8);
+  hipLaunchKernelGGL((gemm_nt32_kernel<SAFE, GROUP>), dim3(grid), dim3(512),
                      131072, stream, A, B, C, M, N, K, swizzle_xcd);
   return hipGetLastError();
 }
@@ -260,12 +508,20 @@ static inline hipError_t launch_gemm_nt(const u16* A, const u16* B, u16* C,
                                         int M, int N, int K,
                                         hipStream_t stream,
                                         int swizzle_xcd = 1, int safe = 0,
-                                        int group = 0) {
-  if (safe) {
-    if (group == 8) return launch_gemm_nt_t<1, 8>(A, B, C, M, N, K, stream, swizzle_xcd);
-    return launch_gemm_nt_t<1, 0>(A, B, C, M, N, K, stream, swizzle_xcd);
+                                        int group = 0, int ph = 2) {
+  if (ph == 32) {
+    if (group == 8) return launch_gemm_nt32_t<0, 8>(A, B, C, M, N, K, stream, swizzle_xcd);
+    return launch_gemm_nt32_t<0, 0>(A, B, C, M, N, K, stream, swizzle_xcd);
   }
-  if (group == 8) return launch_gemm_nt_t<0, 8>(A, B, C, M, N, K, stream, swizzle_xcd);
-  if (group == 16) return launch_gemm_nt_t<0, 16>(A, B, C, M, N, K, stream, swizzle_xcd);
-  return launch_gemm_nt_t<0, 0>(A, B, C, M, N, K, stream, swizzle_xcd);
+  if (safe) {
+    if (group == 8) return launch_gemm_nt_t<1, 8, 4>(A, B, C, M, N, K, stream, swizzle_xcd);
+    return launch_gemm_nt_t<1, 0, 4>(A, B, C, M, N, K, stream, swizzle_xcd);
+  }
+  if (ph == 2) {
+    if (group == 8) return launch_gemm_nt_t<0, 8, 2>(A, B, C, M, N, K, stream, swizzle_xcd);
+    return launch_gemm_nt_t<0, 0, 2>(A, B, C, M, N, K, stream, swizzle_xcd);
+  }
+  if (group == 8) return launch_gemm_nt_t<0, 8, 4>(A, B, C, M, N, K, stream, swizzle_xcd);
+  if (group == 16) return launch_gemm_nt_t<0, 16, 4>(A, B, C, M, N, K, stream, swizzle_xcd);
+  return launch_gemm_nt_t<0, 0, 4>(A, B, C, M, N, K, stream, swizzle_xcd);
 }

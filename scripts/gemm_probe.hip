@@ -32,6 +32,7 @@ int main(int argc, char** argv) {
   int iters = argc > 4 ? atoi(argv[4]) : 20;
   int swz_xcd = argc > 5 ? atoi(argv[5]) : 1;
   int safe = argc > 6 ? atoi(argv[6]) : 0;
+  int ph = argc > 9 ? atoi(argv[9]) : 2;
   int group = argc > 7 ? atoi(argv[7]) : 0;
 
   std::mt19937 rng(1234);
@@ -47,7 +48,7 @@ int main(int argc, char** argv) {
   HC(hipMemcpy(dA, hA.data(), hA.size() * 2, hipMemcpyHostToDevice));
   HC(hipMemcpy(dB, hB.data(), hB.size() * 2, hipMemcpyHostToDevice));
 
-  HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group));
+  HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group, ph));
   HC(hipDeviceSynchronize());
 
   // refcheck: 16 sampled rows, full N, host fp32 accumulate
@@ -82,10 +83,11 @@ int main(int argc, char** argv) {
 
   // determinism: re-run DET times, byte-compare against the first C
   int det = argc > 8 ? atoi(argv[8]) : 3;
+  // argv[9] = phases (2|4)
   std::vector<u16> hC2((size_t)M * N);
   int ndet = 0;
   for (int d = 0; d < det; ++d) {
-    HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group));
+    HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group, ph));
     HC(hipDeviceSynchronize());
     HC(hipMemcpy(hC2.data(), dC, hC2.size() * 2, hipMemcpyDeviceToHost));
     size_t diff = 0;
@@ -99,18 +101,18 @@ int main(int argc, char** argv) {
   HC(hipEventCreate(&t0));
   HC(hipEventCreate(&t1));
   for (int i = 0; i < 3; ++i)
-    HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group));
+    HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group, ph));
   HC(hipDeviceSynchronize());
   HC(hipEventRecord(t0));
   for (int i = 0; i < iters; ++i)
-    HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group));
+    HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group, ph));
   HC(hipEventRecord(t1));
   HC(hipEventSynchronize(t1));
   float ms;
   HC(hipEventElapsedTime(&ms, t0, t1));
   double tf = 2.0 * M * N * K * iters / (ms * 1e-3) / 1e12;
-  printf("gemm_nt %dx%dx%d swz%d safe%d grp%d: %.3f ms/iter  %.1f TF/s\n",
-         M, N, K, swz_xcd, safe, group, ms / iters, tf);
+  printf("gemm_nt %dx%dx%d swz%d safe%d grp%d ph%d: %.3f ms/iter  %.1f TF/s\n",
+         M, N, K, swz_xcd, safe, group, ph, ms / iters, tf);
   printf("PASS\n");
   return 0;
 }
