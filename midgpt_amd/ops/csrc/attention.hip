@@ -550,11 +550,16 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
   const u16* krow = kg + (long)myk * C;
   const u16* vrow = vg + (long)myk * C;
 
-  // wave-owned K row fragments; V re-read per tile (register budget)
-  bf16x8_t kf[NCH];
+  // K row fragments: registers when the budget allows, else re-read from
+  // global (L2-resident) per tile — at C=128 the resident copy pushes the
+  // kernel to 256 VGPR with spills, and the re-read is cheaper (KF_RES).
+  constexpr bool KF_RES = (C <= 64);
+  bf16x8_t kf[KF_RES ? NCH : 1];
+  if (KF_RES) {
 #pragma unroll
-  for (int ch = 0; ch < NCH; ++ch)
-    kf[ch] = *(const bf16x8_t*)(krow + 16 * ch + 8 * (lane >> 5));
+    for (int ch = 0; ch < NCH; ++ch)
+      kf[ch] = *(const bf16x8_t*)(krow + 16 * ch + 8 * (lane >> 5));
+  }
 
   f32x16 dvacc[NCB], dkacc[NCB];
 #pragma unroll
@@ -605,7 +610,9 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int ch = 0; ch < NCH; ++ch) {
         bf16x8_t a = read_rm_frag<C>(ldsQ, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
-        s = mfma_32x32x16_bf16(a, kf[ch], s);
+        bf16x8_t kfr = KF_RES ? kf[ch]
+            : *(const bf16x8_t*)(krow + 16 * ch + 8 * (lane >> 5));
+        s = mfma_32x32x16_bf16(a, kfr, s);
       }
       float p[16];
 #pragma unroll
@@ -621,7 +628,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
       bf16x8_t pf0, pf1;
       if (ABLATE == 4) {  // skip the permlane pack, keep p live
         asm volatile("" :: "v"(p[0]), "v"(p[8]));
-        pf0 = kf[0]; pf1 = kf[1];
+        pf0 = kf[0]; pf1 = kf[0];
       } else {
         pf0 = dlayout_to_afrag(p);
         pf1 = dlayout_to_afrag(p + 8);
@@ -629,7 +636,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int cb = 0; cb < NCB; ++cb) {
         bf16x8_t b0, b1;
-        if (ABLATE == 5) { b0 = kf[0]; b1 = kf[1]; }  // skip LDS frag reads
+        if (ABLATE == 5) { b0 = kf[0]; b1 = kf[0]; }  // skip LDS frag reads
         else {
           b0 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 16 * (lane >> 5));
           b1 = read_tr_frag(ldsDOt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
@@ -653,7 +660,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
       bf16x8_t df0, df1;
       if (ABLATE == 4) {
         asm volatile("" :: "v"(ds[0]), "v"(ds[8]));
-        df0 = kf[0]; df1 = kf[1];
+        df0 = kf[0]; df1 = kf[0];
       } else {
         df0 = dlayout_to_afrag(ds);
         df1 = dlayout_to_afrag(ds + 8);
@@ -661,7 +668,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int cb = 0; cb < NCB; ++cb) {
         bf16x8_t b0, b1;
-        if (ABLATE == 5) { b0 = kf[0]; b1 = kf[1]; }
+        if (ABLATE == 5) { b0 = kf[0]; b1 = kf[0]; }
         else {
           b0 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 16 * (lane >> 5));
           b1 = read_tr_frag(ldsQt, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
