@@ -305,10 +305,18 @@ class ShardedAdamW:
     # ------------------------------------------------------------------
     # per-step
     # ------------------------------------------------------------------
+    def defer_until(self, evt) -> None:
+        """Make the next optimizer step wait on ``evt`` before mutating
+        master/m/v (used by the async checkpoint D2H snapshot)."""
+        self._defer_evt = evt
+
     def step(self, lr: float, g_accum_iters: int = 1) -> torch.Tensor:
         """Apply one optimizer step. Returns a device scalar tensor from
         which the pre-clip global grad norm is ``sqrt(t) * scale`` (no host
         sync on the step path). ``lr`` is this step's scheduled LR."""
+        if getattr(self, "_defer_evt", None) is not None:
+            torch.cuda.current_stream().wait_event(self._defer_evt)
+            self._defer_evt = None
         if self._overlap:
             self._drain()
         scale = 1.0 / (g_accum_iters * self.world)

@@ -58,11 +58,42 @@ class CheckpointManager:
         return self.save_interval > 0 and step > 0 and step % self.save_interval == 0
 
     def save(self, step: int, engine, extra: dict | None = None):
-        """Async save: snapshot to host, write in a background thread."""
+        """Async save: D2H snapshot on a side copy stream (pinned staging),
+        file write in a background thread. The engine's next optimizer
+        step waits on the snapshot event instead of the train loop
+        blocking here."""
         self.wait()
         shard = engine.state_shard()
-        host = {k: (v.detach().to("cpu", copy=True) if torch.is_tensor(v) else v)
-                for k, v in shard.items()}
+        evt = None
+        on_gpu = any(torch.is_tensor(v) and v.is_cuda for v in shard.values())
+        if on_gpu:
+            if not hasattr(self, "_copy_stream"):
+                self._copy_stream = torch.cuda.Stream()
+                self._pinned = {}
+            host = {}
+            with torch.cuda.stream(self._copy_stream):
+                self._copy_stream.wait_stream(torch.cuda.current_stream())
+                for k, v in shard.items():
+                    if torch.is_tensor(v):
+                        buf = self._pinned.get(k)
+                        if buf is None or buf.shape != v.shape or \
+                                buf.dtype != v.dtype:
+                            buf = torch.empty_like(v, device="cpu",
+                                                   pin_memory=True)
+                            self._pinned[k] = buf
+                        buf.copy_(v.detach(), non_blocking=True)
+                        host[k] = buf
+                    else:
+                        host[k] = v
+                evt = torch.cuda.Event()
+                evt.record(self._copy_stream)
+            # master/m/v must not be overwritten before the copy drains:
+            # the engine waits on this event at its next step()
+            engine.defer_until(evt)
+        else:
+            host = {k: (v.detach().to("cpu", copy=True)
+                        if torch.is_tensor(v) else v)
+                    for k, v in shard.items()}
         if extra:
             host["extra"] = extra
         manifest = {
@@ -82,6 +113,8 @@ class CheckpointManager:
         d = ckpt_dir(self.rundir, step)
 
         def _write():
+            if evt is not None:
+                evt.synchronize()  # pinned staging buffers fully landed
             os.makedirs(d, exist_ok=True)
             if write_shard:
                 torch.save(host, os.path.join(d, f"rank{rank:05d}.pt"))

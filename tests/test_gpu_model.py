@@ -225,3 +225,32 @@ def test_training_trajectory_tracks_cpu_reference():
     diffs = [abs(a - b) for a, b in zip(gpu, cpu)]
     assert max(diffs[:10]) < 0.08, (gpu[:10], cpu[:10])
     assert abs(gpu[-1] - cpu[-1]) < 0.3, (gpu[-1], cpu[-1])
+
+
+def test_train_from_real_bin_with_async_checkpoint(tmp_path):
+    """Full train() on GPU from actual uint16 .bin files (pinned-prefetch
+    H2D path + async copy-stream checkpoint snapshot), then resume."""
+    import numpy as np
+    from midgpt_amd.config import ExperimentConfig
+    from midgpt_amd.train import train
+    from midgpt_amd.utils import checkpoint as ckpt
+    rng = np.random.default_rng(0)
+    rng.integers(0, 512, 200_000).astype(np.uint16).tofile(
+        tmp_path / "train.bin")
+    rng.integers(0, 512, 20_000).astype(np.uint16).tofile(
+        tmp_path / "val.bin")
+    cfg = ExperimentConfig(
+        rundir=str(tmp_path / "run"), data_dir=str(tmp_path),
+        learning_rate=1e-3, batch_size=8, warmup_steps=2, min_lr=1e-4,
+        lr_decay_steps=50, max_steps=6, beta2=0.95, weight_decay=1e-4,
+        eval_interval=3, param_dtype="float32", compute_dtype="bfloat16",
+        g_accum_iters=2, shard_model=False, model_config=SMALL, seed=3)
+    train(cfg)
+    step = ckpt.latest_step(cfg.rundir)
+    assert step == 5, step
+    state = ckpt.load_full_state(cfg.rundir)
+    assert torch.isfinite(state["master"]).all()
+    # resume path: continues from the checkpoint without error
+    cfg.max_steps = 8
+    train(cfg)
+    assert ckpt.latest_step(cfg.rundir) == 7
