@@ -310,6 +310,167 @@ __global__ __launch_bounds__(NW * 64, MINW) void attn_fwd_kernel(const u16* __re
 }
 
 // ===========================================================================
+// Forward, DEPTH-2 prefetch variant (SPW=1): four LDS buffers, loads for
+// tile kt+2 issued while tile kt computes — each load gets ~2 compute
+// phases + a barrier to land instead of T14's single phase (the kernels
+// are SQ_WAIT memory-bound at 2 waves/SIMD; this deepens the cover at
+// +~12 VGPR). 4-tile unrolled loop keeps every stage object and buffer
+// index static (rule 20: runtime-indexed vector arrays go to scratch).
+// ===========================================================================
+template <int C, int NW>
+__global__ __launch_bounds__(NW * 64, 2) void attn_fwd_d2_kernel(
+    const u16* __restrict__ q, const u16* __restrict__ k,
+    const u16* __restrict__ v, u16* __restrict__ o, float* __restrict__ lse,
+    int B, int H, int T) {
+  constexpr int NCB = C / 32;
+  constexpr int NCH = C / 16;
+  const float scale = rsqrtf((float)C);
+  const long bh = blockIdx.x % ((long)B * H);
+  const int qb = blockIdx.x / (B * H);
+  const int q0 = qb * (NW * 32);
+  const int lane = lane_id();
+  const int w = wave_id();
+  const int qw0 = q0 + 32 * w;
+  const int myq = qw0 + (lane & 31);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* ldsK = (u16*)smem;                       // [4][32*C]
+  u16* ldsVt = (u16*)(smem + 4 * 32 * C * 2);   // [4][C*32]
+  float* obuf = (float*)smem;                   // epilogue reuse
+
+  const u16* qg = q + (bh * T) * C;
+  const u16* kg = k + (bh * T) * C;
+  const u16* vg = v + (bh * T) * C;
+
+  bf16x8_t qf[NCH];
+  {
+    const u16* qrow = qg + (long)myq * C;
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch)
+      qf[ch] = *(const bf16x8_t*)(qrow + 16 * ch + 8 * (lane >> 5));
+  }
+  f32x16 oacc[NCB];
+#pragma unroll
+  for (int cb = 0; cb < NCB; ++cb) oacc[cb] = (f32x16)(0.f);
+  float m = -1e30f, lsum = 0.f;
+
+  const int nkt = (q0 + NW * 32) / 32;  // multiple of NW (>= 8): % 4 == 0
+
+  auto tile_compute = [&](int kt, const u16* kb, const u16* vb) {
+    const int k0 = kt * 32;
+    if (k0 > qw0 + 31) return;  // wave-uniform
+    f32x16 s = (f32x16)(0.f);
+#pragma unroll
+    for (int ch = 0; ch < NCH; ++ch) {
+      bf16x8_t a = read_rm_frag<C>(kb, lane & 31, 16 * ch * 2 + 16 * (lane >> 5));
+      s = mfma_32x32x16_bf16(a, qf[ch], s);
+    }
+    float sv[16];
+    float mt = -1e30f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      sv[r] = (k0 + mfma_d_row(lane, r) > myq) ? -1e30f : s[r];
+      mt = fmaxf(mt, sv[r]);
+    }
+    mt = fmaxf(mt, __shfl_xor(mt, 32));
+    const float mn = fmaxf(m, mt);
+    const float alpha = __expf((m - mn) * scale);
+    const bool need_rescale = !__all(mt <= m);
+    m = mn;
+    float pp[16], psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      pp[r] = __expf((sv[r] - mn) * scale);
+      psum += pp[r];
+    }
+    psum += __shfl_xor(psum, 32);
+    lsum = lsum * alpha + psum;
+    if (need_rescale) {
+      float arow[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) arow[r] = shfl32(alpha, mfma_d_row(lane, r));
+#pragma unroll
+      for (int cb = 0; cb < NCB; ++cb)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[cb][r] *= arow[r];
+    }
+    bf16x8_t pf0 = dlayout_to_afrag(pp);
+    bf16x8_t pf1 = dlayout_to_afrag(pp + 8);
+#pragma unroll
+    for (int cb = 0; cb < NCB; ++cb) {
+      bf16x8_t b0 = read_tr_frag(vb, 32 * cb + (lane & 31), 16 * (lane >> 5));
+      bf16x8_t b1 = read_tr_frag(vb, 32 * cb + (lane & 31), 32 + 16 * (lane >> 5));
+      oacc[cb] = mfma_32x32x16_bf16(pf0, b0, oacc[cb]);
+      oacc[cb] = mfma_32x32x16_bf16(pf1, b1, oacc[cb]);
+    }
+  };
+
+  // prologue: tile 0 staged synchronously; tile 1's loads issued (stO)
+  stage_rm<C, NW * 64>(kg, ldsK);
+  stage_tr<C, NW * 64>(vg, ldsVt);
+  RmStage<C, NW * 64> kE, kO;
+  TrStage<C, NW * 64> vE, vO;
+  if (1 < nkt) { kO.load(kg + (long)32 * C); vO.load(vg + (long)32 * C); }
+  __syncthreads();
+
+#define FWD_D2_STEP(J, ST_LD, ST_WR, LBUF, WBUF)                              \
+  do {                                                                        \
+    if (kt + (J) + 2 < nkt) {                                                 \
+      k##ST_LD.load(kg + (long)(kt + (J) + 2) * 32 * C);                      \
+      v##ST_LD.load(vg + (long)(kt + (J) + 2) * 32 * C);                      \
+    }                                                                         \
+    tile_compute(kt + (J), ldsK + (LBUF) * 32 * C, ldsVt + (LBUF) * C * 32);  \
+    if (kt + (J) + 1 < nkt) {                                                 \
+      k##ST_WR.write(ldsK + (WBUF) * 32 * C);                                 \
+      v##ST_WR.write_tr(ldsVt + (WBUF) * C * 32);                             \
+    }                                                                         \
+    __syncthreads();                                                          \
+  } while (0)
+
+  for (int kt = 0; kt < nkt; kt += 4) {
+    // invariant at loop top: buf0 = tile kt (ready); stO holds tile kt+1
+    FWD_D2_STEP(0, E, O, 0, 1);  // issue kt+2 (E); compute kt; write kt+1
+    FWD_D2_STEP(1, O, E, 1, 2);  // issue kt+3 (O); compute kt+1; write kt+2
+    FWD_D2_STEP(2, E, O, 2, 3);
+    FWD_D2_STEP(3, O, E, 3, 0);  // write kt+4 into buf0: invariant holds
+  }
+#undef FWD_D2_STEP
+
+  // epilogue: normalize, bounce through LDS, wide stores
+  const float rec = 1.f / lsum;
+  float rrow[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) rrow[r] = shfl32(rec, mfma_d_row(lane, r));
+  if (lane < 32) lse[bh * T + myq] = m * scale + __logf(lsum);
+  float* ob = obuf + w * 32 * 32;
+  u16* og = o + (bh * T + qw0) * C;
+#pragma unroll
+  for (int cb = 0; cb < NCB; ++cb) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = mfma_d_row(lane, r);
+      *(float*)((char*)ob + row * 128 + (((lane & 31) * 4) ^ ((row & 7) << 4))) =
+          oacc[cb][r] * rrow[r];
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    const int row = lane & 31;
+    const int c16 = 16 * (lane >> 5);
+    float tmp[16];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      f32x4 t = *(const f32x4*)((char*)ob + row * 128 + (((c16 + 4 * i) * 4) ^ ((row & 7) << 4)));
+      tmp[4 * i] = t[0]; tmp[4 * i + 1] = t[1]; tmp[4 * i + 2] = t[2]; tmp[4 * i + 3] = t[3];
+    }
+    u16x8 out0, out1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { out0[j] = f2b(tmp[j]); out1[j] = f2b(tmp[8 + j]); }
+    *(u16x8*)(og + (long)row * C + 32 * cb + c16) = out0;
+    *(u16x8*)(og + (long)row * C + 32 * cb + c16 + 8) = out1;
+    __builtin_amdgcn_s_waitcnt(0);
+  }
+}
+
+// ===========================================================================
 // Forward, PAIRED variant: two 32-row KV tiles per barrier round with ONE
 // merged online-softmax rescale — halves barrier rounds and O-rescale
 // passes vs the single-tile loop. nkt is always even (multiple of NW).
@@ -521,8 +682,8 @@ __global__ void attn_delta_kernel(const u16* __restrict__ dO,
 // stage-barrier-compute structure exposes). dK/dV accumulate in registers.
 // S and dS are recomputed from Q,K,LSE (standard flash recompute).
 // ===========================================================================
-template <int C, int NW, int ABLATE = 0>
-__global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
+template <int C, int NW, int ABLATE = 0, int MINW = 2>
+__global__ __launch_bounds__(NW * 64, MINW) void attn_bwd_dkv_kernel(
     const u16* __restrict__ dO, const u16* __restrict__ q,
     const u16* __restrict__ k, const u16* __restrict__ v,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -553,7 +714,7 @@ __global__ __launch_bounds__(NW * 64, 2) void attn_bwd_dkv_kernel(
   // K row fragments: registers when the budget allows, else re-read from
   // global (L2-resident) per tile — at C=128 the resident copy pushes the
   // kernel to 256 VGPR with spills, and the re-read is cheaper (KF_RES).
-  constexpr bool KF_RES = (C <= 64);
+  constexpr bool KF_RES = (C <= 64) && (MINW <= 2);
   bf16x8_t kf[KF_RES ? NCH : 1];
   if (KF_RES) {
 #pragma unroll

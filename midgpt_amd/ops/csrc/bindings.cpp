@@ -206,6 +206,24 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
                      smem, cur_stream(), (const u16*)q.data_ptr(),              \
                      (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),        \
                      (u16*)o.data_ptr(), lse.data_ptr<float>(), B, H, T)
+  // depth-2 prefetch single-tile kernel (4 LDS buffers, loads 2 tiles
+  // ahead): opt-in A/B via MIDGPT_ATTN_FWD_D2=1
+  static const bool fwd_d2 = getenv("MIDGPT_ATTN_FWD_D2") != nullptr;
+  if (fwd_d2 && NW == 8) {
+    size_t smem_d2 = std::max((size_t)(8 * 32 * C * 2), (size_t)(NW * 32 * 32 * 4));
+    if (C == 128)
+      hipLaunchKernelGGL((attn_fwd_d2_kernel<128, 8>), dim3(grid), dim3(512),
+                         smem_d2, cur_stream(), (const u16*)q.data_ptr(),
+                         (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),
+                         (u16*)o.data_ptr(), lse.data_ptr<float>(), B, H, T);
+    else
+      hipLaunchKernelGGL((attn_fwd_d2_kernel<64, 8>), dim3(grid), dim3(512),
+                         smem_d2, cur_stream(), (const u16*)q.data_ptr(),
+                         (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),
+                         (u16*)o.data_ptr(), lse.data_ptr<float>(), B, H, T);
+    launch_check();
+    return {o, lse};
+  }
   // paired-tile kernel (one merged rescale per 64 k): measured +4% at
   // C=128, -4% at C=64 -> default for C=128 only (MIDGPT_ATTN_FWD2=1
   // forces it everywhere, MIDGPT_ATTN_FWD1=1 disables).
@@ -329,6 +347,29 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dO, torch::Tensor q, torch::Te
                        (const u16*)v.data_ptr(), lse.data_ptr<float>(),         \
                        delta.data_ptr<float>(), (u16*)dq.data_ptr(), B, H, T);  \
   } while (0)
+  static const bool bwd_minw3 = getenv("MIDGPT_ATTN_BWD_MINW3") != nullptr;
+  if (C == 64 && NW_A == 8 && bwd_minw3 && !getenv("MIDGPT_DKV_ABLATE")) {
+    // squeeze to 3 waves/SIMD (K frags re-read from L2; compiler caps
+    // VGPRs at 168) — A/B experiment for the 124M config
+    if (smem_a > 64 * 1024)
+      hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&attn_bwd_dkv_kernel<64, 8, 0, 3>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem_a);
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 8, 0, 3>), dim3(grid_a),
+                       dim3(512), smem_a, cur_stream(),
+                       (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),
+                       (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (u16*)dk.data_ptr(), (u16*)dv.data_ptr(), B, H, T);
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<64, 8, 1, 3>), dim3(grid_b),
+                       dim3(512), smem_b, cur_stream(),
+                       (const u16*)dO.data_ptr(), (const u16*)q.data_ptr(),
+                       (const u16*)k.data_ptr(), (const u16*)v.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (u16*)dq.data_ptr(), B, H, T);
+    launch_check();
+    return {dq, dk, dv};
+  }
   if (C == 128 && NW_A == 8) LAUNCH_BWD(128, 8, 8, 1, 2);
   else if (C == 128) LAUNCH_BWD(128, 4, 4, 1, 2);
   else if (NW_B == 8) LAUNCH_BWD(64, 8, 8, 1, 2);
