@@ -170,6 +170,14 @@ def train(config: ExperimentConfig):
             loss_step = one_step()
 
         if it % 20 == 0:
+            # global-batch mean at log points (C3: the reference's logged
+            # loss is the cross-device mean; collective only when logging)
+            if world > 1:
+                t = torch.tensor([loss_step], dtype=torch.float64,
+                                 device=device if device.type == "cuda"
+                                 else "cpu")
+                pdist.all_reduce_(t)
+                loss_step = float(t[0]) / world
             log_metrics(config, it, {"loss/optimized": loss_step})
         if mngr is not None and mngr.should_save(it):
             mngr.save(it, engine)
