@@ -88,7 +88,10 @@ DEVINL void stage_half(const u16* __restrict__ g, long rows0, long ldb,
 // barrier) — used to bisect schedule races from layout bugs.
 // ORDER: 0 = column-major tile walk (B-panel reuse), 1 = grouped walk
 // (GROUP tile_m rows per super-column: A panels L2-resident too).
-template <int SAFE, int GROUP, int PH>
+// ABL=1: skip all in-loop ds_reads (fragments loaded once, kept live —
+// measures the MFMA+stage+barrier structure's ceiling). ABL=2: also skip
+// the stage glds (pure MFMA+barrier ceiling). Timing-only (wrong math).
+template <int SAFE, int GROUP, int PH, int ABL = 0>
 __launch_bounds__(512, 1)
 __global__ void gemm_nt_kernel(const u16* __restrict__ A,
                                const u16* __restrict__ B,
@@ -157,13 +160,17 @@ __global__ void gemm_nt_kernel(const u16* __restrict__ A,
 
 // slot: afr register slot; the fragment is rows (q*32 + (slot&1)*16)
 #define READ_AP(pp, q, ks, slot)                                              \
-  afr[slot][ks] = *(const bf16x8g*)(lds + LDS_A(pp) + wave_m * 16384 +        \
-      swz((q) * 32 + ((slot) & 1) * 16 + a_row, (ks) * 64 + frag_colb))
+  do { if (ABL == 0)                                                          \
+    afr[slot][ks] = *(const bf16x8g*)(lds + LDS_A(pp) + wave_m * 16384 +      \
+        swz((q) * 32 + ((slot) & 1) * 16 + a_row, (ks) * 64 + frag_colb));    \
+  } while (0)
 #define READ_A(q, ks, slot) READ_AP(par, q, ks, slot)
 #define READ_BP(pp, nr, ks)                                                   \
-  bfr[nr][ks] = *(const bf16x8g*)(lds + LDS_B(pp) +                           \
-      ((b_nloc + (nr) * 16) >> 7) * 16384 +                                   \
-      swz((b_nloc + (nr) * 16) & 127, (ks) * 64 + frag_colb))
+  do { if (ABL == 0)                                                          \
+    bfr[nr][ks] = *(const bf16x8g*)(lds + LDS_B(pp) +                         \
+        ((b_nloc + (nr) * 16) >> 7) * 16384 +                                 \
+        swz((b_nloc + (nr) * 16) & 127, (ks) * 64 + frag_colb));              \
+  } while (0)
 #define READ_B(nr, ks) READ_BP(par, nr, ks)
 #define MFMA_QUAD2(q, base)                                                   \
   _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                            \
@@ -181,9 +188,35 @@ __global__ void gemm_nt_kernel(const u16* __restrict__ A,
                   __builtin_amdgcn_s_barrier();                               \
                   asm volatile("" ::: "memory"); } while (0)
 #define PRIO(x) __builtin_amdgcn_s_setprio(x)
+// compile-time interleave of the post-MFMA prefetch reads into the MFMA
+// cluster (T19): without it hipcc front-loads the ds_reads, serializing
+// the read segment against the MFMA segment (measured: reads = +25%% of
+// kernel time at zero overlap). Masks: MFMA=0x8, DS_READ=0x100.
+#define SGB(mask, n) __builtin_amdgcn_sched_group_barrier(mask, n, 0)
+#define INTERLEAVE(nds)                                                       \
+  _Pragma("unroll") for (int gg = 0; gg < 4; ++gg) {                          \
+    SGB(0x8, 4); SGB(0x100, nds);                                             \
+  }
+// variant with the phase's stage glds (VMEM 0x10) woven in as well
+#define INTERLEAVE2(nds)                                                      \
+  _Pragma("unroll") for (int gg = 0; gg < 4; ++gg) {                          \
+    SGB(0x8, 2); SGB(0x100, nds); SGB(0x8, 2); SGB(0x10, 1);                  \
+  }
 #define VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
 
-  if constexpr (PH == 4) {
+  if constexpr (ABL > 0) {
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        bfr[nr][ks] = *(const bf16x8g*)(lds + LDS_B(0) + nr * 2048 + ks * 64);
+#pragma unroll
+    for (int sl = 0; sl < 4; ++sl)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        afr[sl][ks] = *(const bf16x8g*)(lds + LDS_A(0) + sl * 2048 + ks * 64);
+  }
+  if constexpr (PH == 4 && ABL == 0) {
     // tile 0's fragments (the loop reads tile u+1's at u.p3)
 #pragma unroll
     for (int nr = 0; nr < 4; ++nr) { READ_BP(0, nr, 0); READ_BP(0, nr, 1); }
@@ -245,33 +278,40 @@ __global__ void gemm_nt_kernel(const u16* __restrict__ A,
       continue;
     }
 
+#define STAGE_IF(cond, ...) do { if (ABL < 2 && (cond)) stage_half(__VA_ARGS__); } while (0)
     // ---- 4 phases per tile, POST-MFMA fragment prefetch: phase q's
     // MFMA consumes afr pair (q&1); the reads for phase q+1 are issued
     // right after the MFMA cluster into the other pair, so their LDS
     // latency hides under the barrier + next phase's stage segment.
     // B(u+1)+strip0(u+1) reads go after p3's vmcnt(4)+MFMA (guarded).
     // phase 0: stage A-lo(u+1); MFMA q0 [afr pair0]; read strip1->pair1
-    if (u + 1 < NT) stage_half(A, bm + 0, ldab, kb1, lds + LDS_A(nxt) + 0);
-    BAR();
-    PRIO(1); MFMA_QUAD2(0, 0); PRIO(0);
+    // ONE barrier per phase: every phase's reads/stages now live inside
+    // its MFMA region, and all slot-reuse distances span >= 2 regions, so
+    // the second barrier of the classic template guards nothing.
+    PRIO(1);
+    MFMA_QUAD2(0, 0);
     READ_A(1, 0, 2); READ_A(1, 1, 2); READ_A(1, 0, 3); READ_A(1, 1, 3);
+    STAGE_IF(u + 1 < NT, A, bm + 0, ldab, kb1, lds + LDS_A(nxt) + 0);
+    INTERLEAVE2(1);
+    PRIO(0);
     BAR();
-    // phase 1: stage A-hi(u+1); MFMA q1 [pair1]; read strip2->pair0
-    if (u + 1 < NT) stage_half(A, bm + 128, ldab, kb1, lds + LDS_A(nxt) + 16384);
-    BAR();
-    PRIO(1); MFMA_QUAD2(1, 2); PRIO(0);
+    // phase 1
+    PRIO(1);
+    MFMA_QUAD2(1, 2);
     READ_A(2, 0, 0); READ_A(2, 1, 0); READ_A(2, 0, 1); READ_A(2, 1, 1);
+    STAGE_IF(u + 1 < NT, A, bm + 128, ldab, kb1, lds + LDS_A(nxt) + 16384);
+    INTERLEAVE2(1);
+    PRIO(0);
     BAR();
-    // phase 2: stage B-lo(u+2); MFMA q2 [pair0]; read strip3->pair1
-    if (u + 2 < NT) stage_half(B, bn + 0, ldab, kb2, lds + LDS_B(par) + 0);
-    BAR();
-    PRIO(1); MFMA_QUAD2(2, 0); PRIO(0);
+    // phase 2: both B(u+2) halves staged here, then the tile guard
+    PRIO(1);
+    MFMA_QUAD2(2, 0);
     READ_A(3, 0, 2); READ_A(3, 1, 2); READ_A(3, 0, 3); READ_A(3, 1, 3);
-    BAR();
-    // phase 3: stage B-hi(u+2); vmcnt; MFMA q3 [pair1]; read B(u+1) +
-    // strip0(u+1) -> pair0 (safe: A(u+1)/B(u+1) landed per the vmcnt)
-    if (u + 2 < NT) {
-      stage_half(B, bn + 128, ldab, kb2, lds + LDS_B(par) + 16384);
+    STAGE_IF(u + 2 < NT, B, bn + 0, ldab, kb2, lds + LDS_B(par) + 0);
+    STAGE_IF(u + 2 < NT, B, bn + 128, ldab, kb2, lds + LDS_B(par) + 16384);
+    INTERLEAVE2(1);
+    PRIO(0);
+    if (ABL < 2 && u + 2 < NT) {
       VMCNT(4);  // leaves exactly B(u+2)'s 4 glds; A(u+1)/B(u+1) landed
     } else {
       // tail: B(u+2) skipped, so the 4 newest in-flight glds would be
@@ -279,13 +319,20 @@ __global__ void gemm_nt_kernel(const u16* __restrict__ A,
       VMCNT(0);
     }
     BAR();
-    PRIO(1); MFMA_QUAD2(3, 2); PRIO(0);
-    if (u + 1 < NT) {
+    // phase 3: MFMA q3 + next tile's fragments (guarded by the vmcnt+bar)
+    PRIO(1);
+    MFMA_QUAD2(3, 2);
+    {
+      // unconditional (single basic block, so the interleave directives
+      // can act); at the last tile read this tile's slots again (dead)
+      const int snxt = (u + 1 < NT) ? nxt : par;
 #pragma unroll
-      for (int nr = 0; nr < 4; ++nr) { READ_BP(nxt, nr, 0); READ_BP(nxt, nr, 1); }
-      READ_AP(nxt, 0, 0, 0); READ_AP(nxt, 0, 1, 0);
-      READ_AP(nxt, 0, 0, 1); READ_AP(nxt, 0, 1, 1);
+      for (int nr = 0; nr < 4; ++nr) { READ_BP(snxt, nr, 0); READ_BP(snxt, nr, 1); }
+      READ_AP(snxt, 0, 0, 0); READ_AP(snxt, 0, 1, 0);
+      READ_AP(snxt, 0, 0, 1); READ_AP(snxt, 0, 1, 1);
+      INTERLEAVE(3);
     }
+    PRIO(0);
     BAR();
   }
 #undef READ_A
@@ -308,6 +355,183 @@ __global__ void gemm_nt_kernel(const u16* __restrict__ A,
   }
 }
 
+
+// ---------------------------------------------------------------------
+// 1-wave/SIMD variant: 4 waves (256 threads), each owning a 128x128
+// sub-tile — accumulators spill into the unified AGPR file (gfx950:
+// 512 regs/lane at 1 wave/SIMD). Motivation (measured): at 8 waves the
+// MFMA skeleton runs at the issue floor but the per-tile LDS fragment
+// reads (24 per 64 MFMAs per wave) add ~47% that never hides; this
+// shape needs only 32 reads per 128 MFMAs (0.25/MFMA) and has no
+// co-resident wave competing for the SIMD.
+// ---------------------------------------------------------------------
+DEVINL void stage_half4(const u16* __restrict__ g, long rows0, long ldb,
+                        long kbyte0, char* lds_base) {
+  const int w = wave_id();
+  const int l = lane_id();
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int q = (w * 4 + j) * 1024 + l * 16;
+    int row = q >> 7;
+    int colb = (q & 127) ^ ((row & 7) << 4);
+    const char* src = (const char*)g + (rows0 + row) * ldb + kbyte0 + colb;
+    GLDS(src, lds_base + (w * 4 + j) * 1024);
+  }
+}
+
+template <int GROUP>
+__launch_bounds__(256, 1)
+__global__ void gemm_nt_1w_kernel(const u16* __restrict__ A,
+                                  const u16* __restrict__ B,
+                                  u16* __restrict__ C,
+                                  int M, int N, int K, int swizzle_xcd) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int l = lane_id();
+  const int w = wave_id();
+  const int wave_m = w >> 1;       // 0..1 (128-row half)
+  const int wave_n = w & 1;        // 0..1 (128-col half)
+  const int tiles_m = M >> 8;
+  const int tiles_n = N >> 8;
+
+  int bid = blockIdx.x;
+  if (swizzle_xcd) {
+    const int nwg = tiles_m * tiles_n;
+    const int qq = nwg >> 3, rr = nwg & 7;
+    const int xcd = bid & 7, idx = bid >> 3;
+    bid = (xcd < rr ? xcd * (qq + 1) : rr * (qq + 1) + (xcd - rr) * qq) + idx;
+  }
+  long bm, bn;
+  if (GROUP > 0) {
+    const int band = GROUP * tiles_n;
+    const int b0 = bid / band, r0 = bid % band;
+    bm = (long)(b0 * GROUP + r0 % GROUP) << 8;
+    bn = (long)(r0 / GROUP) << 8;
+  } else {
+    bm = (long)(bid % tiles_m) << 8;
+    bn = (long)(bid / tiles_m) << 8;
+  }
+
+  const long ldab = (long)K * 2;
+  const int NT = K >> 6;
+
+  f32x4 acc[8][8] = {};   // 256 regs -> AGPR half of the unified file
+
+  stage_half4(A, bm + 0, ldab, 0, smem + LDS_A(0) + 0);
+  stage_half4(A, bm + 128, ldab, 0, smem + LDS_A(0) + 16384);
+  stage_half4(B, bn + 0, ldab, 0, smem + LDS_B(0) + 0);
+  stage_half4(B, bn + 128, ldab, 0, smem + LDS_B(0) + 16384);
+  if (NT > 1) {
+    stage_half4(B, bn + 0, ldab, 128, smem + LDS_B(1) + 0);
+    stage_half4(B, bn + 128, ldab, 128, smem + LDS_B(1) + 16384);
+  }
+  // first 16 glds (A(0)+B(0)) must land; B(1)'s 8 may fly
+  if (NT == 1) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  const int frag_colb = (l >> 4) * 16;
+  const int a_row = l & 15;
+  const int b_nloc0 = wave_n * 128 + (l & 15);
+  char* const lds = smem;
+
+  bf16x8g bfr[8][2];   // B frags [nr 0..7][ks]
+  bf16x8g afr[4][2];   // two pairs (double-buffer across phases)
+
+#define RD1_A(pp, q, ks, slot)                                                \
+  afr[slot][ks] = *(const bf16x8g*)(lds + LDS_A(pp) + wave_m * 16384 +        \
+      swz((q) * 32 + ((slot) & 1) * 16 + a_row, (ks) * 64 + frag_colb))
+#define RD1_B(pp, nr, ks)                                                     \
+  bfr[nr][ks] = *(const bf16x8g*)(lds + LDS_B(pp) +                           \
+      ((b_nloc0 + (nr) * 16) >> 7) * 16384 +                                  \
+      swz((b_nloc0 + (nr) * 16) & 127, (ks) * 64 + frag_colb))
+#define MFMA_Q1(q, base)                                                      \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                            \
+    _Pragma("unroll") for (int mr = 0; mr < 2; ++mr)                          \
+      _Pragma("unroll") for (int nr = 0; nr < 8; ++nr)                        \
+        acc[(q) * 2 + mr][nr] = mfma16(afr[(base) + mr][ks], bfr[nr][ks],     \
+                                       acc[(q) * 2 + mr][nr])
+#define BAR1() do { asm volatile("" ::: "memory");                            \
+                  __builtin_amdgcn_s_barrier();                               \
+                  asm volatile("" ::: "memory"); } while (0)
+#define ILV1(nds)                                                             \
+  _Pragma("unroll") for (int gg = 0; gg < 8; ++gg) {                          \
+    SGB(0x8, 2); SGB(0x100, nds); SGB(0x8, 2); SGB(0x10, 1);                  \
+  }
+
+  // tile 0 fragments
+#pragma unroll
+  for (int nr = 0; nr < 8; ++nr) { RD1_B(0, nr, 0); RD1_B(0, nr, 1); }
+  RD1_A(0, 0, 0, 0); RD1_A(0, 0, 1, 0); RD1_A(0, 0, 0, 1); RD1_A(0, 0, 1, 1);
+
+  for (int u = 0; u < NT; ++u) {
+    const int par = u & 1;
+    const int nxt = par ^ 1;
+    const long kb1 = (long)(u + 1) << 7;
+    const long kb2 = (long)(u + 2) << 7;
+
+    // phase 0: 32 MFMA (q0) + strip1 reads + stage A-lo(u+1)
+    PRIO(1);
+    MFMA_Q1(0, 0);
+    RD1_A(par, 1, 0, 2); RD1_A(par, 1, 1, 2); RD1_A(par, 1, 0, 3); RD1_A(par, 1, 1, 3);
+    if (u + 1 < NT) stage_half4(A, bm + 0, ldab, kb1, lds + LDS_A(nxt) + 0);
+    ILV1(1);
+    PRIO(0);
+    BAR1();
+    // phase 1
+    PRIO(1);
+    MFMA_Q1(1, 2);
+    RD1_A(par, 2, 0, 0); RD1_A(par, 2, 1, 0); RD1_A(par, 2, 0, 1); RD1_A(par, 2, 1, 1);
+    if (u + 1 < NT) stage_half4(A, bm + 128, ldab, kb1, lds + LDS_A(nxt) + 16384);
+    ILV1(1);
+    PRIO(0);
+    BAR1();
+    // phase 2: B(u+2) both halves + the tile guard
+    PRIO(1);
+    MFMA_Q1(2, 0);
+    RD1_A(par, 3, 0, 2); RD1_A(par, 3, 1, 2); RD1_A(par, 3, 0, 3); RD1_A(par, 3, 1, 3);
+    if (u + 2 < NT) {
+      stage_half4(B, bn + 0, ldab, kb2, lds + LDS_B(par) + 0);
+      stage_half4(B, bn + 128, ldab, kb2, lds + LDS_B(par) + 16384);
+    }
+    ILV1(1);
+    PRIO(0);
+    if (u + 2 < NT) { VMCNT(8); } else { VMCNT(0); }
+    BAR1();
+    // phase 3: q3 + next tile's fragments
+    PRIO(1);
+    MFMA_Q1(3, 2);
+    {
+      const int snxt = (u + 1 < NT) ? nxt : par;
+#pragma unroll
+      for (int nr = 0; nr < 8; ++nr) { RD1_B(snxt, nr, 0); RD1_B(snxt, nr, 1); }
+      RD1_A(snxt, 0, 0, 0); RD1_A(snxt, 0, 1, 0);
+      RD1_A(snxt, 0, 0, 1); RD1_A(snxt, 0, 1, 1);
+      ILV1(3);
+    }
+    PRIO(0);
+    BAR1();
+  }
+#undef RD1_A
+#undef RD1_B
+#undef MFMA_Q1
+#undef ILV1
+
+  // epilogue: scalar bf16 stores (as the 8-wave kernel)
+  const long ldc = N;
+  const long crow0 = bm + wave_m * 128 + (l >> 4) * 4;
+  const long ccol0 = bn + wave_n * 128 + (l & 15);
+#pragma unroll
+  for (int mr = 0; mr < 8; ++mr) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long row = crow0 + mr * 16 + r;
+#pragma unroll
+      for (int nr = 0; nr < 8; ++nr)
+        C[row * ldc + ccol0 + nr * 16] = f2b(acc[mr][nr][r]);
+    }
+  }
+}
 
 // ---------------------------------------------------------------------
 // 32x32x16-MFMA sibling: same tile/LDS/staging/schedule, half the MFMA
@@ -466,7 +690,7 @@ __global__ void gemm_nt32_kernel(const u16* __restrict__ A,
 }
 
 // host-side launcher (shared by bindings and the standalone probe)
-template <int SAFE, int GROUP, int PH>
+template <int SAFE, int GROUP, int PH, int ABL = 0>
 static hipError_t launch_gemm_nt_t(const u16* A, const u16* B, u16* C,
                                    int M, int N, int K, hipStream_t stream,
                                    int swizzle_xcd) {
@@ -475,13 +699,13 @@ static hipError_t launch_gemm_nt_t(const u16* A, const u16* B, u16* C,
   static int lds_set = 0;
   if (!lds_set) {
     (void)hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&gemm_nt_kernel<SAFE, GROUP, PH>),
+        reinterpret_cast<const void*>(&gemm_nt_kernel<SAFE, GROUP, PH, ABL>),
         hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
     lds_set = 1;
   }
   const int grid = (M >> 8) * (N >> 8);
-  hipLaunchKernelGGL((gemm_nt_kernel<SAFE, GROUP, PH>), dim3(grid), dim3(512),
-                     131072, stream, A, B, C, M, N, K, swizzle_xcd);
+  hipLaunchKernelGGL((gemm_nt_kernel<SAFE, GROUP, PH, ABL>), dim3(grid),
+                     dim3(512), 131072, stream, A, B, C, M, N, K, swizzle_xcd);
   return hipGetLastError();
 }
 
@@ -517,6 +741,23 @@ static inline hipError_t launch_gemm_nt(const u16* A, const u16* B, u16* C,
     if (group == 8) return launch_gemm_nt_t<1, 8, 4>(A, B, C, M, N, K, stream, swizzle_xcd);
     return launch_gemm_nt_t<1, 0, 4>(A, B, C, M, N, K, stream, swizzle_xcd);
   }
+  if (ph == 10) {
+    if (M % 256 || N % 256 || K % 64) return hipErrorInvalidValue;
+    static int lds1 = 0;
+    if (!lds1) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&gemm_nt_1w_kernel<8>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+      lds1 = 1;
+    }
+    hipLaunchKernelGGL((gemm_nt_1w_kernel<8>), dim3((M >> 8) * (N >> 8)),
+                       dim3(256), 131072, stream, A, B, C, M, N, K, swizzle_xcd);
+    return hipGetLastError();
+  }
+  if (ph == 101)
+    return launch_gemm_nt_t<0, 8, 4, 1>(A, B, C, M, N, K, stream, swizzle_xcd);
+  if (ph == 102)
+    return launch_gemm_nt_t<0, 8, 4, 2>(A, B, C, M, N, K, stream, swizzle_xcd);
   if (ph == 2) {
     if (group == 8) return launch_gemm_nt_t<0, 8, 2>(A, B, C, M, N, K, stream, swizzle_xcd);
     return launch_gemm_nt_t<0, 0, 2>(A, B, C, M, N, K, stream, swizzle_xcd);

@@ -51,11 +51,13 @@ int main(int argc, char** argv) {
   HC(launch_gemm_nt(dA, dB, dC, M, N, K, 0, swz_xcd, safe, group, ph));
   HC(hipDeviceSynchronize());
 
+  const bool abl = ph >= 100;  // timing-only structure ablations
   // refcheck: 16 sampled rows, full N, host fp32 accumulate
   std::vector<u16> hC((size_t)M * N);
   HC(hipMemcpy(hC.data(), dC, hC.size() * 2, hipMemcpyDeviceToHost));
   double max_rel = 0, max_abs = 0;
   int bad = 0;
+  if (!abl) {
   std::uniform_int_distribution<int> rowd(0, M - 1);
   for (int s = 0; s < 16; ++s) {
     int m = rowd(rng);
@@ -80,9 +82,11 @@ int main(int argc, char** argv) {
   printf("refcheck: max_rel %.3e max_abs %.3e bad %d (16 rows x all cols)\n",
          max_rel, max_abs, bad);
   if (bad > 0) { printf("FAIL\n"); return 1; }
+  }
 
   // determinism: re-run DET times, byte-compare against the first C
-  int det = argc > 8 ? atoi(argv[8]) : 3;
+  {
+  int det = abl ? 0 : (argc > 8 ? atoi(argv[8]) : 3);
   // argv[9] = phases (2|4)
   std::vector<u16> hC2((size_t)M * N);
   int ndet = 0;
@@ -95,7 +99,7 @@ int main(int argc, char** argv) {
     if (diff) { ++ndet; printf("  nondet run %d: %zu elems differ\n", d, diff); }
   }
   if (ndet) { printf("NONDETERMINISTIC (%d/%d runs)\nFAIL\n", ndet, det); return 1; }
-
+  }
   // timing
   hipEvent_t t0, t1;
   HC(hipEventCreate(&t0));
