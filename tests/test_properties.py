@@ -88,3 +88,35 @@ def test_grad_clip_bounds_update_norm(seed):
     # b1=b2=0 -> m = g_clipped; mhat = m; vhat = g^2; update = sign-ish...
     # check the clipped gradient's norm directly via m
     assert float(m.norm()) <= 1.0 + 1e-5
+
+
+@given(total=st.integers(1, 5000), world=st.integers(1, 8),
+       nb=st.integers(1, 6))
+@settings(max_examples=60, deadline=None)
+def test_bucket_piece_map_partitions_buffer(total, world, nb):
+    """The bucketed ZeRO piece maps must tile the padded buffer exactly:
+    every rank's pieces are disjoint, bucket-aligned, and the union over
+    ranks covers [0, padded) (checkpoint reassembly correctness)."""
+    shard = (total + world - 1) // world
+    padded = shard * world
+    nb_eff = max(1, min(nb, shard))
+    cuts = [shard * i // nb_eff for i in range(nb_eff + 1)]
+    buckets = [(cuts[i] * world, (cuts[i + 1] - cuts[i]) * world)
+               for i in range(nb_eff)]
+    assert sum(n for _, n in buckets) == padded
+    covered = []
+    for rank in range(world):
+        po = 0
+        for (o_b, n_b) in buckets:
+            p_b = n_b // world
+            covered.append((o_b + rank * p_b, p_b))
+            po += p_b
+        assert po == shard
+    covered.sort()
+    pos = 0
+    for off, n in covered:
+        if n == 0:
+            continue
+        assert off == pos, (off, pos)
+        pos += n
+    assert pos == padded
