@@ -16,23 +16,24 @@
 //   the swizzle is applied to the per-lane SOURCE address and to the
 //   ds_read address (both-sides involution, guide §5.4 rule 21).
 //
-// Schedule per K-tile u (2 raw s_barriers per phase, counted vmcnt only —
-// never __syncthreads, which would drain the in-flight glds):
-//   p0: ds_read B(u) [8xb128] + A strip0 [4]; glds A-lo(u+1)
-//   p1: ds_read A strip1;                     glds A-hi(u+1)
-//   p2: ds_read A strip2;                     glds B-lo(u+2)
-//   p3: ds_read A strip3;                     glds B-hi(u+2); vmcnt(4)
-//   The p3 vmcnt(4)+barrier is the guard for tile u+1's reads at the
-//   next p0 (A(u+1) = the 2 oldest of the 4 newest in-flight halves;
-//   B(u+1) landed earlier); the prologue ends with its own vmcnt(4) +
-//   barrier guarding tile 0.
-//   (each phase: ... ; s_barrier ; setprio(1) ; 16 MFMA ; setprio(0) ;
-//    s_barrier)
-// Slot-lifetime: A(u+1) overwrites A(u-1) (dead after (u-1).p3 barrier);
-// B(u+2) overwrites B(u) (dead after u.p0 barrier). vmcnt(4) at p3
-// guarantees A(u+1) landed before (u+1).p0's reads; vmcnt(6) at p0
-// guarantees A(u)+B(u) landed on the first tile (prologue stages 6
-// halves: A(0), B(0), B(1)).
+// Schedule per K-tile u — ONE raw s_barrier per phase, counted vmcnt
+// only (never __syncthreads, which would drain the in-flight glds).
+// Each phase is a single scheduling region {setprio(1); 16 MFMA; next
+// phase's fragment ds_reads; this phase's stage glds; setprio(0)} with
+// sched_group_barrier directives weaving the reads/stages INTO the MFMA
+// stream (without them hipcc front-loads the reads and the read segment
+// runs fully serial to the MFMA segment — measured +26% kernel time):
+//   p0: MFMA q0 | read A strip1      | glds A-lo(u+1)
+//   p1: MFMA q1 | read A strip2      | glds A-hi(u+1)
+//   p2: MFMA q2 | read A strip3      | glds B-lo(u+2)+B-hi(u+2); vmcnt(4)
+//   p3: MFMA q3 | read B(u+1)+A strip0(u+1) (guarded by p2's vmcnt+bar)
+// Slot-lifetime: A(u+1) overwrites A(u-1) (last read 2 regions back);
+// B(u+2) overwrites B(u) (last read 3 regions back). The p2 vmcnt(4)
+// leaves only B(u+2)'s glds in flight, so A(u+1)/B(u+1) have landed
+// before p3's prefetch reads; at the loop tail (stages skipped) the
+// wait degrades to vmcnt(0) because the 4 newest in-flight loads would
+// otherwise be the very tile about to be read. The prologue stages
+// A(0), B(0), B(1) and guards tile 0 with vmcnt(4) (= B(1) in flight).
 //
 // Operand layouts ("NT"): A[M,K] row-major, B[N,K] row-major, C[M,N]
 // row-major = A @ B^T — the natural layout of torch Linear fwd

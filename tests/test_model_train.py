@@ -297,3 +297,13 @@ def test_generate_beyond_block_size():
         logits = model(cur[:, -16:])
         cur = torch.cat([cur, logits[:, -1].argmax(-1)[:, None]], dim=1)
     assert torch.equal(out, cur)
+
+
+def test_device_prefetcher_cpu_passthrough():
+    from midgpt_amd.utils.prefetch import DevicePrefetcher
+    pre = DevicePrefetcher(torch.device("cpu"))
+    x = torch.arange(12).reshape(3, 4)
+    y = x + 1
+    h = pre.start(x, y)
+    xd, yd = pre.wait(h)
+    assert torch.equal(xd, x) and torch.equal(yd, y)
