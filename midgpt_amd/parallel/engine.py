@@ -148,6 +148,11 @@ class ShardedAdamW:
         if self._overlap:
             self._comm_stream = torch.cuda.Stream(device=device)
             self._pending = []  # [(done_event, buf_index), ...]
+        if self.world > 1:
+            # force communicator init with a default-stream collective NOW:
+            # the first real collective fires from a backward hook on the
+            # side stream, which must never be the lazy-rendezvous path
+            pdist.all_reduce_(torch.zeros(1, device=device))
         # ---- backward-hook bucket machinery (GPU only) ------------------
         # bucket b covers flat range [o_b, o_b+n_b); a param contributes to
         # every bucket its range overlaps. Buckets launch in descending
