@@ -711,10 +711,11 @@ __global__ __launch_bounds__(NW * 64, MINW) void attn_bwd_dkv_kernel(
   const u16* krow = kg + (long)myk * C;
   const u16* vrow = vg + (long)myk * C;
 
-  // K row fragments: registers when the budget allows, else re-read from
-  // global (L2-resident) per tile — at C=128 the resident copy pushes the
-  // kernel to 256 VGPR with spills, and the re-read is cheaper (KF_RES).
-  constexpr bool KF_RES = (C <= 64) && (MINW <= 2);
+  // K row fragments resident in registers (the L2 re-read variant was
+  // measured -20% on dkv C=128 in a full-step profile despite removing
+  // the 2-VGPR spill — the per-tile global loads add latency the spill
+  // never cost); re-read only under the MINW=3 squeeze experiment.
+  constexpr bool KF_RES = (MINW <= 2);
   bf16x8_t kf[KF_RES ? NCH : 1];
   if (KF_RES) {
 #pragma unroll
