@@ -24,12 +24,16 @@ def main():
     p.add_argument("--steps", type=int, default=3000)
     p.add_argument("--out", default="gpurun_out/losscurve")
     p.add_argument("--data", default="data/local_char")
+    p.add_argument("--config", default="shakespeare_char")
+    p.add_argument("--batch", type=int, default=None)
     p.add_argument("--debug", action="store_true",
                    help="1-batch evals (CPU smoke)")
     args = p.parse_args()
 
-    config = load_config("shakespeare_char")
+    config = load_config(args.config)
     config.data_dir = args.data
+    if args.batch:
+        config.batch_size = args.batch
     config.model_config.vocab_size = 98  # local_char alphabet
     config.max_steps = args.steps
     config.eval_interval = max(1, args.steps // 12)
