@@ -34,9 +34,10 @@ def main():
     config.data_dir = args.data
     if args.batch:
         config.batch_size = args.batch
+        config.g_accum_iters = 1
     config.model_config.vocab_size = 98  # local_char alphabet
     config.max_steps = args.steps
-    config.eval_interval = max(1, args.steps // 12)
+    config.eval_interval = max(1, args.steps // (12 if args.steps >= 1000 else 6))
     config.rundir = args.out
     config.seed = 1234
     config.debug = args.debug
