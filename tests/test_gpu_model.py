@@ -1,6 +1,5 @@
 """GPU end-to-end: model forward/backward on HIP kernels vs CPU reference;
 one engine step; extension-presence guard."""
-import math
 
 import pytest
 import torch

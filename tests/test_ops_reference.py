@@ -1,8 +1,6 @@
 """CPU numerics tests for the reference op layer (the oracle the HIP
 kernels are tested against) and the autograd dispatch wrappers."""
-import math
 
-import pytest
 import torch
 
 from midgpt_amd import ops

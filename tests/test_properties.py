@@ -1,8 +1,6 @@
 """Property-based tests for the op layer's mathematical invariants (CPU,
 using the same dispatch wrappers the model uses)."""
-import math
 
-import pytest
 import torch
 from hypothesis import given, settings, strategies as st
 
