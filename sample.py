@@ -47,19 +47,13 @@ def load_model(ckpt_dir: str, device: torch.device) -> tuple[GPT, ExperimentConf
     return model, config
 
 
-def main():
-    p = argparse.ArgumentParser()
-    p.add_argument("--ckpt_dir", required=True)
-    p.add_argument("--start", default="\n")
-    p.add_argument("--num_samples", type=int, default=3)
-    p.add_argument("--max_new_tokens", type=int, default=200)
-    p.add_argument("--temperature", type=float, default=0.8)
-    p.add_argument("--seed", type=int, default=None)
-    args = p.parse_args()
-
-    device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
-    model, config = load_model(args.ckpt_dir, device)
-
+def load_model_and_tokenizer(ckpt_dir: str, device):
+    """Model + encode/decode fns for a rundir (tokenizer: char meta.pkl
+    if the config's data_dir has one, else tiktoken GPT-2 — reference
+    sample.py:143-159)."""
+    device = torch.device(device) if not isinstance(device, torch.device) \
+        else device
+    model, config = load_model(ckpt_dir, device)
     meta_path = os.path.join(config.data_dir, "meta.pkl")
     if os.path.exists(meta_path):
         with open(meta_path, "rb") as f:
@@ -71,6 +65,22 @@ def main():
         enc = tiktoken.get_encoding("gpt2")
         encode = lambda s: enc.encode(s, allowed_special={"<|endoftext|>"})  # noqa: E731
         decode = enc.decode
+    return model, encode, decode, config
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--ckpt_dir", required=True)
+    p.add_argument("--start", default="\n")
+    p.add_argument("--num_samples", type=int, default=3)
+    p.add_argument("--max_new_tokens", type=int, default=200)
+    p.add_argument("--temperature", type=float, default=0.8)
+    p.add_argument("--seed", type=int, default=None)
+    args = p.parse_args()
+
+    device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
+    model, encode, decode, config = load_model_and_tokenizer(args.ckpt_dir,
+                                                             device)
 
     gen = torch.Generator(device="cpu")
     if args.seed is not None:

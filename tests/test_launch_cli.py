@@ -66,3 +66,44 @@ def test_launch_and_sample_cli(tmp_path):
         assert "---------------" in r3.stdout
     finally:
         os.remove(cfg_path)
+
+
+def test_serve_endpoint(tmp_path):
+    """serve.py: health + generation over a tiny trained rundir."""
+    import json
+    import pickle
+
+    import numpy as np
+
+    from midgpt_amd.train import train
+    from tests.test_model_train import tiny_config
+
+    # char "dataset" metadata so the server picks the char tokenizer
+    data_dir = tmp_path / "data"
+    data_dir.mkdir()
+    alphabet = [chr(97 + i) for i in range(37)]
+    with open(data_dir / "meta.pkl", "wb") as f:
+        pickle.dump({"vocab_size": 37,
+                     "stoi": {c: i for i, c in enumerate(alphabet)},
+                     "itos": {i: c for i, c in enumerate(alphabet)}}, f)
+    cfg = tiny_config(tmp_path / "run", max_steps=2, eval_interval=10,
+                      data_dir=str(data_dir))
+    train(cfg)
+    # launch.py normally freezes config.json; do it here for the loader
+    with open(tmp_path / "run" / "config.json", "w") as f:
+        f.write(cfg.to_json())
+
+    from starlette.testclient import TestClient
+
+    from serve import build_app
+    app = build_app(str(tmp_path / "run"), device="cpu")
+    client = TestClient(app)
+    r = client.get("/healthz")
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+    r = client.post("/generate", json={"prompt": "ab", "max_new_tokens": 5,
+                                       "num_samples": 2, "seed": 0,
+                                       "temperature": 1.0})
+    assert r.status_code == 200
+    samples = r.json()["samples"]
+    assert len(samples) == 2
+    assert all(s.startswith("ab") and len(s) == 7 for s in samples)
