@@ -117,8 +117,7 @@ class MLP(nn.Module):
                     and os.environ.get("MIDGPT_FUSED_MLP") == "1"):
                 return self.dropout(ops.fused_mlp(x, self.c_fc.weight,
                                                   self.c_proj.weight))
-            return self.dropout(
-                self.c_proj(F.gelu(self.c_fc(x), approximate="tanh")))
+            return self.dropout(self.c_proj(ops.gelu(self.c_fc(x))))
 
 
 class Block(nn.Module):

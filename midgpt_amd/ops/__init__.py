@@ -239,6 +239,36 @@ def flash_attention(q, k, v):
 
 
 # ----------------------------------------------------------------------------
+# GELU (tanh approx) fwd/bwd (reference src/model.py:30; plan K7): a
+# hand-written u16x8 elementwise pair replacing the torch library kernels
+# (same numerics as F.gelu(approximate="tanh"); fp32 internal).
+# ----------------------------------------------------------------------------
+class _Gelu(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.save_for_backward(x)
+        if _use_hip(x) and x.dtype == torch.bfloat16 and x.numel() % 8 == 0:
+            return _C.gelu_fwd(x.contiguous())
+        return torch.nn.functional.gelu(x, approximate="tanh")
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        if _use_hip(x) and x.dtype == torch.bfloat16 and x.numel() % 8 == 0:
+            return _C.gelu_bwd(dy.contiguous(), x.contiguous())
+        xf = x.float()
+        c0, c1 = 0.7978845608028654, 0.044715
+        inner = c0 * (xf + c1 * xf ** 3)
+        t = torch.tanh(inner)
+        dg = 0.5 * (1 + t) + 0.5 * xf * (1 - t * t) * c0 * (1 + 3 * c1 * xf ** 2)
+        return (dy.float() * dg).to(x.dtype)
+
+
+def gelu(x):
+    return _Gelu.apply(x)
+
+
+# ----------------------------------------------------------------------------
 # Embedding: gather fwd, scatter-add bwd (reference src/layers.py:13-34;
 # plan K8). The forward gather is a plain coalesced index_select (torch);
 # the backward is the hand-written HIP fp32-accurate scatter-add.

@@ -403,6 +403,32 @@ torch::Tensor embedding_bwd(torch::Tensor dy, torch::Tensor idx, long V) {
 }
 
 // ---------------------------------------------------------------------------
+torch::Tensor gelu_fwd(torch::Tensor x) {
+  CHECK_GPU(x);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.numel() % 8 == 0,
+              "gelu: bf16 with numel %% 8 == 0");
+  auto y = torch::empty_like(x);
+  long n = x.numel();
+  long grid = std::min((n / 8 + 255) / 256 + 1, (long)8192);
+  hipLaunchKernelGGL(gelu_fwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const u16*)x.data_ptr(), (u16*)y.data_ptr(), n);
+  launch_check();
+  return y;
+}
+
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x) {
+  CHECK_GPU(dy); CHECK_GPU(x);
+  auto dx = torch::empty_like(x);
+  long n = x.numel();
+  long grid = std::min((n / 8 + 255) / 256 + 1, (long)8192);
+  hipLaunchKernelGGL(gelu_bwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const u16*)dy.data_ptr(), (const u16*)x.data_ptr(),
+                     (u16*)dx.data_ptr(), n);
+  launch_check();
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
 torch::Tensor probe_mfma(torch::Tensor A, torch::Tensor B) {
   CHECK_GPU(A); CHECK_GPU(B);
   auto D = torch::zeros({32, 32}, A.options());
@@ -433,6 +459,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_bwd", &attn_bwd);
   mod.def("embedding_bwd", &embedding_bwd);
+  mod.def("gelu_fwd", &gelu_fwd);
+  mod.def("gelu_bwd", &gelu_bwd);
   mod.def("linear_gelu_fwd", &linear_gelu_fwd);
   mod.def("lt_probe", &lt_probe);
   mod.def("matmul_dgelu", &matmul_dgelu);

@@ -6,5 +6,6 @@
 #include "adamw.hip"
 #include "attention.hip"
 #include "embedding.hip"
+#include "gelu.hip"
 #include "probe.hip"
 #include "bindings.cpp"

@@ -245,3 +245,18 @@ def test_attention_fwd_spiked_max():
     o_ref = torch.matmul(torch.softmax(s, -1), v.float())
     assert relerr(lse, lse_ref) < 1e-3
     assert relerr(o, o_ref) < 3e-2, relerr(o, o_ref)
+
+
+def test_gelu_fwd_bwd():
+    """HIP tanh-GELU pair vs torch fp32 reference (K7)."""
+    torch.manual_seed(10)
+    x = (torch.randn(4096, 512, device=DEV) * 3).to(torch.bfloat16)
+    x.requires_grad_(True)
+    y = ops.gelu(x)
+    x2 = x.detach().float().requires_grad_(True)
+    y2 = torch.nn.functional.gelu(x2, approximate="tanh")
+    assert relerr(y, y2) < 1e-2
+    g = torch.randn_like(y)
+    (y.float() * g.float()).sum().backward()
+    (y2 * g.float()).sum().backward()
+    assert relerr(x.grad, x2.grad) < 2e-2
