@@ -410,8 +410,14 @@ torch::Tensor gelu_fwd(torch::Tensor x) {
   auto y = torch::empty_like(x);
   long n = x.numel();
   long grid = std::min((n / 8 + 255) / 256 + 1, (long)8192);
-  hipLaunchKernelGGL(gelu_fwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+  static const int gnt = getenv("MIDGPT_GELU_NT0") ? 0 : 1;
+  if (gnt)
+    hipLaunchKernelGGL((gelu_fwd_bf16<1>), dim3(grid), dim3(256), 0, cur_stream(),
                      (const u16*)x.data_ptr(), (u16*)y.data_ptr(), n);
+  else
+    hipLaunchKernelGGL((gelu_fwd_bf16<0>), dim3(grid), dim3(256), 0,
+                       cur_stream(), (const u16*)x.data_ptr(),
+                       (u16*)y.data_ptr(), n);
   launch_check();
   return y;
 }
@@ -421,9 +427,15 @@ torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x) {
   auto dx = torch::empty_like(x);
   long n = x.numel();
   long grid = std::min((n / 8 + 255) / 256 + 1, (long)8192);
-  hipLaunchKernelGGL(gelu_bwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+  static const int gnt2 = getenv("MIDGPT_GELU_NT0") ? 0 : 1;
+  if (gnt2)
+    hipLaunchKernelGGL((gelu_bwd_bf16<1>), dim3(grid), dim3(256), 0, cur_stream(),
                      (const u16*)dy.data_ptr(), (const u16*)x.data_ptr(),
                      (u16*)dx.data_ptr(), n);
+  else
+    hipLaunchKernelGGL((gelu_bwd_bf16<0>), dim3(grid), dim3(256), 0,
+                       cur_stream(), (const u16*)dy.data_ptr(),
+                       (const u16*)x.data_ptr(), (u16*)dx.data_ptr(), n);
   launch_check();
   return dx;
 }

@@ -28,31 +28,41 @@ DEVINL float dgelu_f(float x) {
   return 0.5f * (1.f + t) + 0.5f * x * (1.f - t * t) * dinner;
 }
 
+// NT: nontemporal loads/stores (1 GB streams have no reuse — keep them
+// out of L2); UNROLL: independent chunks per iteration for load-latency
+// overlap within a wave.
+template <int NT>
 __global__ void gelu_fwd_bf16(const u16* __restrict__ x, u16* __restrict__ y,
                               long n) {
   long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   const long step = (long)gridDim.x * blockDim.x * 8;
   for (; i + 7 < n; i += step) {
-    u16x8 v = *(const u16x8*)(x + i);
+    u16x8 v = NT ? __builtin_nontemporal_load((const u16x8*)(x + i))
+                 : *(const u16x8*)(x + i);
     u16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) o[j] = f2b(gelu_f(b2f(v[j])));
-    *(u16x8*)(y + i) = o;
+    if (NT) __builtin_nontemporal_store(o, (u16x8*)(y + i));
+    else *(u16x8*)(y + i) = o;
   }
 }
 
+template <int NT>
 __global__ void gelu_bwd_bf16(const u16* __restrict__ dy,
                               const u16* __restrict__ x,
                               u16* __restrict__ dx, long n) {
   long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   const long step = (long)gridDim.x * blockDim.x * 8;
   for (; i + 7 < n; i += step) {
-    u16x8 g = *(const u16x8*)(dy + i);
-    u16x8 v = *(const u16x8*)(x + i);
+    u16x8 g = NT ? __builtin_nontemporal_load((const u16x8*)(dy + i))
+                 : *(const u16x8*)(dy + i);
+    u16x8 v = NT ? __builtin_nontemporal_load((const u16x8*)(x + i))
+                 : *(const u16x8*)(x + i);
     u16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j)
       o[j] = f2b(b2f(g[j]) * dgelu_f(b2f(v[j])));
-    *(u16x8*)(dx + i) = o;
+    if (NT) __builtin_nontemporal_store(o, (u16x8*)(dx + i));
+    else *(u16x8*)(dx + i) = o;
   }
 }
