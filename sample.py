@@ -61,7 +61,13 @@ def load_model_and_tokenizer(ckpt_dir: str, device):
         encode = lambda s: [meta["stoi"][c] for c in s]  # noqa: E731
         decode = lambda t: "".join(meta["itos"][i] for i in t)  # noqa: E731
     else:
-        import tiktoken
+        try:
+            import tiktoken
+        except ImportError as e:
+            raise SystemExit(
+                "no meta.pkl in the config's data_dir and tiktoken is not "
+                "installed (offline image): token-id I/O only — rerun with "
+                "a char-level dataset or install tiktoken") from e
         enc = tiktoken.get_encoding("gpt2")
         encode = lambda s: enc.encode(s, allowed_special={"<|endoftext|>"})  # noqa: E731
         decode = enc.decode

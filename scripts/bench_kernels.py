@@ -41,6 +41,22 @@ def bench_attn(B=32, H=16, T=1024, C=128):
           f"{flops_bwd/t_bwd/1e12:7.1f} TF/s (causal-effective)")
 
 
+def bench_gelu(N=131072, D=8192):
+    x = torch.randn(N, D, device=DEV).to(torch.bfloat16)
+    dy = torch.randn_like(x)
+    t = timeit(lambda: ops._C.gelu_fwd(x))
+    print(f"gelu_fwd N{N} D{D}: {t*1e3:8.3f} ms  {2*N*D*2/t/1e12:6.2f} TB/s")
+    t = timeit(lambda: ops._C.gelu_bwd(dy, x))
+    print(f"gelu_bwd N{N} D{D}: {t*1e3:8.3f} ms  {3*N*D*2/t/1e12:6.2f} TB/s")
+
+
+def bench_embedding(N=131072, V=50304, D=2048):
+    idx = torch.randint(0, V, (N,), device=DEV)
+    dy = torch.randn(N, D, device=DEV).to(torch.bfloat16)
+    t = timeit(lambda: ops._C.embedding_bwd(dy, idx, V), iters=10)
+    print(f"embedding_bwd N{N} V{V} D{D}: {t*1e3:8.3f} ms")
+
+
 def bench_rmsnorm(N=131072, D=2048):
     x = torch.randn(N, D, device=DEV, dtype=torch.bfloat16)
     y, invr = ops._C.rmsnorm_fwd(x, None, 1e-6)
@@ -110,6 +126,11 @@ if __name__ == "__main__":
     if w in ("all", "attn"):
         bench_attn()
         bench_attn(B=32, H=12, T=1024, C=64)
+        bench_attn(B=4, H=32, T=4096, C=128)  # 7B shape
+    if w in ("all", "gelu"):
+        bench_gelu()
+    if w in ("all", "embed"):
+        bench_embedding()
     if w in ("all", "norm"):
         bench_rmsnorm()
         bench_qkv_prep()
