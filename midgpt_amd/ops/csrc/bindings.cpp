@@ -387,6 +387,23 @@ torch::Tensor embedding_bwd(torch::Tensor dy, torch::Tensor idx, long V) {
   long N = dy.size(0);
   int D = dy.size(1);
   TORCH_CHECK(D % 8 == 0, "embedding_bwd: D % 8 == 0");
+  static const bool use_atomic = getenv("MIDGPT_EMBED_ATOMIC") != nullptr;
+  if (!use_atomic && D % 64 == 0) {
+    // sort-based run-per-wave accumulation (exact fp32, no atomics)
+    auto sorted_perm = idx.sort();
+    auto sorted = std::get<0>(sorted_perm).contiguous();
+    auto perm = std::get<1>(sorted_perm).contiguous();
+    auto dw = torch::zeros({V, (long)D}, dy.options());
+    const int wpb = 4;  // waves per block
+    long grid = (N + wpb - 1) / wpb;
+    for (int d0 = 0; d0 < D; d0 += 2048)
+      hipLaunchKernelGGL(embed_bwd_sorted_kernel, dim3(grid), dim3(wpb * 64),
+                         0, cur_stream(), (const u16*)dy.data_ptr(),
+                         sorted.data_ptr<long>(), perm.data_ptr<long>(),
+                         (u16*)dw.data_ptr(), N, D, d0);
+    launch_check();
+    return dw;
+  }
   auto dw32 = torch::zeros({V, (long)D}, dy.options().dtype(torch::kFloat));
   long grid = std::min(N, (long)4096);
   hipLaunchKernelGGL(embed_bwd_scatter_kernel, dim3(grid), dim3(256), 0,
