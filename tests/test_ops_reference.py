@@ -164,3 +164,21 @@ def test_adamw_step_matches_optax_chain_semantics():
         upd = mhat / (vhat.sqrt() + eps) + wd_over_peak * master2
         master2 = master2 - lr * upd
         assert torch.allclose(master, master2, atol=1e-6)
+
+
+def test_gelu_reference_path_matches_torch():
+    """ops.gelu CPU fallback (fwd + hand-derived bwd) vs torch autograd."""
+    import torch
+
+    from midgpt_amd import ops
+    torch.manual_seed(0)
+    x = (torch.randn(64, 33) * 3).requires_grad_(True)
+    y = ops.gelu(x)
+    x2 = x.detach().clone().requires_grad_(True)
+    y2 = torch.nn.functional.gelu(x2, approximate="tanh")
+    assert torch.allclose(y, y2, atol=1e-6)
+    g = torch.randn_like(y)
+    (y * g).sum().backward()
+    (y2 * g).sum().backward()
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5), \
+        (x.grad - x2.grad).abs().max()
