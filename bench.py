@@ -50,9 +50,10 @@ def main():
     p.add_argument("--local-batch", type=int, default=128,
                    help="sequences per GPU per optimizer step")
     p.add_argument("--micro-batch", type=int, default=None,
-                   help="microbatch size (default: 64 at T<=1024, else 4 — "
-                        "measured: 7B@4096 no-remat micro=4 fits 224 GB and "
-                        "beats remat micro=8 by 15%%)")
+                   help="microbatch size (default: the whole local batch "
+                        "for <=1024-dim models, 64 for larger at T<=1024, "
+                        "4 at long seq — all measured optima; 7B@4096 "
+                        "no-remat micro=4 fits 224 GB)")
     p.add_argument("--remat", action="store_true",
                    help="per-block activation recompute (default OFF: 288 GB "
                         "HBM fits stored activations at the default "
