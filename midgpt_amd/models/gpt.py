@@ -117,7 +117,14 @@ class MLP(nn.Module):
                     and os.environ.get("MIDGPT_FUSED_MLP") == "1"):
                 return self.dropout(ops.fused_mlp(x, self.c_fc.weight,
                                                   self.c_proj.weight))
-            return self.dropout(self.c_proj(ops.gelu(self.c_fc(x))))
+            # torch's gelu kernels win IN-STEP (the hand-written pair ties
+            # them in isolation, but its nontemporal stores defeat the
+            # L2 handoff to the next GEMM — rocprof: 136 vs 109 ms/4
+            # steps at xl). MIDGPT_GELU_HIP=1 opts into the native pair.
+            if os.environ.get("MIDGPT_GELU_HIP") == "1":
+                return self.dropout(self.c_proj(ops.gelu(self.c_fc(x))))
+            return self.dropout(
+                self.c_proj(F.gelu(self.c_fc(x), approximate="tanh")))
 
 
 class Block(nn.Module):

@@ -427,7 +427,7 @@ torch::Tensor gelu_fwd(torch::Tensor x) {
   auto y = torch::empty_like(x);
   long n = x.numel();
   long grid = std::min((n / 8 + 255) / 256 + 1, (long)8192);
-  static const int gnt = getenv("MIDGPT_GELU_NT0") ? 0 : 1;
+  static const int gnt = getenv("MIDGPT_GELU_NT") ? 1 : 0;
   if (gnt)
     hipLaunchKernelGGL((gelu_fwd_bf16<1>), dim3(grid), dim3(256), 0, cur_stream(),
                      (const u16*)x.data_ptr(), (u16*)y.data_ptr(), n);
@@ -444,7 +444,7 @@ torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x) {
   auto dx = torch::empty_like(x);
   long n = x.numel();
   long grid = std::min((n / 8 + 255) / 256 + 1, (long)8192);
-  static const int gnt2 = getenv("MIDGPT_GELU_NT0") ? 0 : 1;
+  static const int gnt2 = getenv("MIDGPT_GELU_NT") ? 1 : 0;
   if (gnt2)
     hipLaunchKernelGGL((gelu_bwd_bf16<1>), dim3(grid), dim3(256), 0, cur_stream(),
                      (const u16*)dy.data_ptr(), (const u16*)x.data_ptr(),
