@@ -107,3 +107,34 @@ def test_serve_endpoint(tmp_path):
     samples = r.json()["samples"]
     assert len(samples) == 2
     assert all(s.startswith("ab") and len(s) == 7 for s in samples)
+
+
+def test_distributed_flag_requires_torchrun():
+    """--distributed without WORLD_SIZE must fail with a clear error."""
+    import subprocess
+    import sys
+    env = dict(os.environ)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "launch.py", "--config", "shakespeare_char",
+         "--distributed"],
+        capture_output=True, text=True, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode != 0
+    assert "torchrun" in r.stderr
+
+
+def test_broadcast_str_single_process():
+    from midgpt_amd.parallel.dist import broadcast_str
+    assert broadcast_str("runs/abc") == "runs/abc"
+
+
+def test_prefetcher_shape_change():
+    import torch
+
+    from midgpt_amd.utils.prefetch import DevicePrefetcher
+    pre = DevicePrefetcher(torch.device("cpu"))
+    for shape in [(2, 4), (3, 5), (2, 4)]:
+        x = torch.ones(shape)
+        xd, yd = pre.wait(pre.start(x, x + 1))
+        assert xd.shape == torch.Size(shape)
