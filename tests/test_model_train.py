@@ -307,3 +307,40 @@ def test_device_prefetcher_cpu_passthrough():
     h = pre.start(x, y)
     xd, yd = pre.wait(h)
     assert torch.equal(xd, x) and torch.equal(yd, y)
+
+
+def test_engine_manifest_matches_named_parameters():
+    """Checkpoint manifest order/offsets == named_parameters traversal
+    (the reference's flat-leaves contract, src/train.py:215)."""
+    model = GPT(TINY)
+    engine = ShardedAdamW(model, compute_dtype=torch.float32, zero=False)
+    man = engine.named_param_manifest()
+    off = 0
+    for entry, (name, p) in zip(man, model.named_parameters()):
+        assert entry["name"] == name
+        assert entry["offset"] == off
+        assert entry["numel"] == p.numel()
+        assert tuple(entry["shape"]) == tuple(p.shape)
+        off += p.numel()
+    assert off == engine.total
+
+
+def test_config_from_json_ignores_unknown_fields():
+    """Forward compatibility: a rundir config.json written by a newer
+    version (extra fields) must still load."""
+    c = load_config("openwebtext_xl")
+    import json as _json
+    d = _json.loads(c.to_json())
+    d["future_field"] = 123
+    d["model_config"]["future_model_field"] = "x"
+    c2 = ExperimentConfig.from_json(_json.dumps(d))
+    assert c2.model_config.n_embd == c.model_config.n_embd
+
+
+def test_zero_flag_collapses_at_world_one():
+    """shard_model=True at world 1 must behave as the replicated engine
+    (full master, no padding surprises)."""
+    model = GPT(TINY)
+    engine = ShardedAdamW(model, compute_dtype=torch.float32, zero=True)
+    assert not engine.zero  # collapses: zero needs world > 1
+    assert engine.master.numel() == engine.total
