@@ -344,3 +344,22 @@ def test_zero_flag_collapses_at_world_one():
     engine = ShardedAdamW(model, compute_dtype=torch.float32, zero=True)
     assert not engine.zero  # collapses: zero needs world > 1
     assert engine.master.numel() == engine.total
+
+
+def test_prepare_shakespeare_offline(tmp_path, monkeypatch):
+    """The char data-prep pipeline runs offline when input.txt exists
+    (reference data/shakespeare_char/prepare.py contract)."""
+    import pickle
+    import sys
+    (tmp_path / "input.txt").write_text("hello world\n" * 200)
+    monkeypatch.setattr(sys, "argv", ["prep", "--out", str(tmp_path)])
+    from midgpt_amd.data_prep import prepare_shakespeare
+    prepare_shakespeare.main()
+    meta = pickle.load(open(tmp_path / "meta.pkl", "rb"))
+    train = np.fromfile(tmp_path / "train.bin", dtype=np.uint16)
+    val = np.fromfile(tmp_path / "val.bin", dtype=np.uint16)
+    assert meta["vocab_size"] == len(set("hello world\n"))
+    assert len(train) + len(val) == 12 * 200
+    # decode round-trip
+    text = "".join(meta["itos"][i] for i in train[:12])
+    assert text == "hello world\n"
